@@ -1,0 +1,99 @@
+# SPDX-License-Identifier: Apache-2.0
+"""In-tree native builds.
+
+``build_cpu()``  — g++  → legate_sparse/_cpu_kernels.so  (C++/OpenMP)
+``build_hip()``  — hipcc --offload-arch=gfx950 → legate_sparse/_hip_kernels.so
+
+Both .so files live in-tree (git-ignored) so they travel with the gpurun
+snapshot.  The HIP extension is pure pybind11 + HIP runtime — no ATen/torch
+link — so it builds anywhere hipcc exists and loads anywhere libamdhip64
+exists; kernels take raw device pointers + the torch stream handle.
+"""
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+import sysconfig
+
+PKG_DIR = os.path.dirname(os.path.abspath(__file__))
+REPO_DIR = os.path.dirname(PKG_DIR)
+SRC_DIR = os.path.join(REPO_DIR, "src")
+
+
+def _includes() -> list:
+    import pybind11
+
+    return [
+        "-I" + pybind11.get_include(),
+        "-I" + sysconfig.get_paths()["include"],
+    ]
+
+
+def _newer(target: str, sources: list) -> bool:
+    if not os.path.exists(target):
+        return False
+    t = os.path.getmtime(target)
+    return all(os.path.getmtime(s) <= t for s in sources)
+
+
+def build_cpu(force: bool = False) -> str:
+    src = [os.path.join(SRC_DIR, "cpu", "cpu_kernels.cpp")]
+    out = os.path.join(PKG_DIR, "_cpu_kernels.so")
+    if not force and _newer(out, src):
+        return out
+    cmd = (
+        ["g++", "-O3", "-march=native", "-std=c++17", "-shared", "-fPIC",
+         "-fopenmp", "-fvisibility=hidden"]
+        + _includes()
+        + src
+        + ["-o", out]
+    )
+    subprocess.run(cmd, check=True)
+    return out
+
+
+HIP_SOURCES = [
+    os.path.join(SRC_DIR, "hip", "module.cpp"),
+    os.path.join(SRC_DIR, "hip", "spmv.hip"),
+    os.path.join(SRC_DIR, "hip", "spgemm.hip"),
+    os.path.join(SRC_DIR, "hip", "convert.hip"),
+    os.path.join(SRC_DIR, "hip", "solver_ops.hip"),
+]
+
+
+def build_hip(force: bool = False) -> str:
+    out = os.path.join(PKG_DIR, "_hip_kernels.so")
+    srcs = [s for s in HIP_SOURCES if os.path.exists(s)]
+    if not srcs:
+        raise RuntimeError("no HIP sources found under src/hip")
+    if not force and _newer(out, srcs):
+        return out
+    hipcc = os.environ.get("HIPCC", "hipcc")
+    cmd = (
+        [hipcc, "--offload-arch=gfx950", "-O3", "-std=c++17", "-shared",
+         "-fPIC", "-fvisibility=hidden", "-Wno-unused-result"]
+        + _includes()
+        + ["-x", "hip"]
+        + srcs
+        + ["-o", out]
+    )
+    subprocess.run(cmd, check=True)
+    return out
+
+
+def build_all(force: bool = False):
+    built = [build_cpu(force)]
+    try:
+        built.append(build_hip(force))
+    except (FileNotFoundError, RuntimeError, subprocess.CalledProcessError) as e:
+        print(f"[legate_sparse build] HIP build skipped/failed: {e}",
+              file=sys.stderr)
+        raise
+    return built
+
+
+if __name__ == "__main__":
+    force = "--force" in sys.argv
+    for p in build_all(force):
+        print("built", p)

@@ -1,0 +1,149 @@
+# SPDX-License-Identifier: Apache-2.0
+"""Shared compressed-format base class.
+
+Counterpart of the reference's ``legate_sparse/base.py`` (CompressedBase,
+base.py:57-250): nnz bookkeeping, astype/copy plumbing, the zero-preserving
+unary ufunc forwarding (base.py:209-250), and ``sum`` via ones-matvec
+(base.py:111-171).  The Legion pos-rect packing (nnz_to_pos/ZIP_TO_RECT1)
+has no counterpart here: we keep plain ``indptr`` tensors, which is the
+MI355X-native replacement for pos stores.
+"""
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from . import comm
+from .runtime import runtime
+from .types import to_torch_dtype, to_numpy_dtype
+
+# Zero-preserving unary ufuncs forwarded to .data
+# (reference base.py:209-250: applying f with f(0)=0 elementwise to a sparse
+# matrix only needs to touch stored values).
+_UNARY_UFUNCS = {
+    "sin": torch.sin,
+    "tan": torch.tan,
+    "arcsin": torch.asin,
+    "arctan": torch.atan,
+    "sinh": torch.sinh,
+    "tanh": torch.tanh,
+    "arcsinh": torch.asinh,
+    "arctanh": torch.atanh,
+    "sqrt": torch.sqrt,
+    "ceil": torch.ceil,
+    "floor": torch.floor,
+    "rint": torch.round,
+    "sign": torch.sign,
+    "expm1": torch.expm1,
+    "log1p": torch.log1p,
+    "deg2rad": torch.deg2rad,
+    "rad2deg": torch.rad2deg,
+    "trunc": torch.trunc,
+}
+
+
+class CompressedBase:
+    """Base for csr_array / dia_array: dtype bookkeeping + ufuncs."""
+
+    # -- subclass contract: _with_data(new_data) -> same-structure copy ----
+
+    @property
+    def dtype(self):
+        return to_numpy_dtype(self._torch_dtype())
+
+    @property
+    def ndim(self) -> int:
+        return 2
+
+    def astype(self, dtype, casting: str = "unsafe", copy: bool = True):
+        tdt = to_torch_dtype(dtype)
+        if not copy and tdt == self._torch_dtype():
+            return self
+        return self._with_data(self._data_tensor().to(tdt))
+
+    def copy(self):
+        return self._with_data(self._data_tensor().clone())
+
+    def conj(self, copy: bool = True):
+        d = self._data_tensor()
+        if not d.is_complex():
+            return self.copy() if copy else self
+        return self._with_data(torch.conj(d).resolve_conj())
+
+    conjugate = conj
+
+    def __neg__(self):
+        return self._with_data(-self._data_tensor())
+
+    # -- ufunc forwarding -------------------------------------------------
+    def __array_ufunc__(self, ufunc, method, *inputs, **kwargs):
+        name = getattr(ufunc, "__name__", None)
+        if method == "__call__" and name in _UNARY_UFUNCS and len(inputs) == 1:
+            return self._with_data(_UNARY_UFUNCS[name](self._data_tensor()))
+        if method == "__call__" and name == "multiply" and len(inputs) == 2:
+            other = inputs[0] if inputs[1] is self else inputs[1]
+            return self.multiply(other)
+        return NotImplemented
+
+    def _forward_ufunc(self, name):
+        return self._with_data(_UNARY_UFUNCS[name](self._data_tensor()))
+
+    # generate named methods: A.sin(), A.sqrt(), ... (scipy-compatible)
+    def sin(self): return self._forward_ufunc("sin")
+    def tan(self): return self._forward_ufunc("tan")
+    def arcsin(self): return self._forward_ufunc("arcsin")
+    def arctan(self): return self._forward_ufunc("arctan")
+    def sinh(self): return self._forward_ufunc("sinh")
+    def tanh(self): return self._forward_ufunc("tanh")
+    def arcsinh(self): return self._forward_ufunc("arcsinh")
+    def arctanh(self): return self._forward_ufunc("arctanh")
+    def sqrt(self): return self._forward_ufunc("sqrt")
+    def ceil(self): return self._forward_ufunc("ceil")
+    def floor(self): return self._forward_ufunc("floor")
+    def rint(self): return self._forward_ufunc("rint")
+    def sign(self): return self._forward_ufunc("sign")
+    def expm1(self): return self._forward_ufunc("expm1")
+    def log1p(self): return self._forward_ufunc("log1p")
+    def deg2rad(self): return self._forward_ufunc("deg2rad")
+    def rad2deg(self): return self._forward_ufunc("rad2deg")
+    def trunc(self): return self._forward_ufunc("trunc")
+
+    # -- helpers -----------------------------------------------------------
+    def _torch_dtype(self) -> torch.dtype:
+        return self._data_tensor().dtype
+
+    def _data_tensor(self) -> torch.Tensor:
+        raise NotImplementedError
+
+    def _with_data(self, data: torch.Tensor):
+        raise NotImplementedError
+
+    def asformat(self, format, copy: bool = False):
+        if format is None or format == self.format:
+            return self.copy() if copy else self
+        conv = getattr(self, "to" + format, None)
+        if conv is None:
+            raise ValueError(f"Format {format} is unknown.")
+        return conv()
+
+
+def as_torch_1d(x, dtype=None, device=None) -> torch.Tensor:
+    """Coerce numpy/torch/sequence input to a 1-D contiguous torch tensor."""
+    if isinstance(x, torch.Tensor):
+        t = x
+    elif isinstance(x, np.ndarray):
+        t = torch.from_numpy(np.ascontiguousarray(x))
+    else:
+        t = torch.as_tensor(x)
+    if dtype is not None:
+        t = t.to(to_torch_dtype(dtype))
+    if device is not None:
+        t = t.to(device)
+    return t.contiguous().reshape(-1)
+
+
+def global_sum_scalar(local: torch.Tensor) -> torch.Tensor:
+    """All-reduce a 1-element tensor (kept on device)."""
+    if runtime.world_size > 1:
+        comm.allreduce_(local)
+    return local

@@ -1,0 +1,44 @@
+# SPDX-License-Identifier: Apache-2.0
+"""API-coverage wrapper (reference legate_sparse/coverage.py:59-107).
+
+``clone_module(scipy.sparse, globals())`` re-exports every public name of
+scipy.sparse that this package does not implement itself, so user code that
+touches an uncovered corner of the scipy API still runs (on host scipy).
+Implemented names are wrapped with a lightweight provenance label used by
+the profiler hooks (utils.annotate).
+"""
+from __future__ import annotations
+
+import functools
+import types
+from typing import Any, Dict
+
+import torch
+
+_PROFILE_RANGES = False  # flipped by utils.enable_profiler_ranges()
+
+
+def track_provenance(fn, name: str = None):
+    label = name or getattr(fn, "__qualname__", str(fn))
+
+    @functools.wraps(fn)
+    def wrapper(*args, **kwargs):
+        if _PROFILE_RANGES and torch.cuda.is_available():
+            torch.cuda.nvtx.range_push(f"legate_sparse::{label}")
+            try:
+                return fn(*args, **kwargs)
+            finally:
+                torch.cuda.nvtx.range_pop()
+        return fn(*args, **kwargs)
+
+    return wrapper
+
+
+def clone_module(src_module: types.ModuleType, dst_globals: Dict[str, Any]):
+    """Fill ``dst_globals`` with names from ``src_module`` that are not
+    already defined — the scipy fall-through of the reference
+    (coverage.py:59-84)."""
+    for name in dir(src_module):
+        if name.startswith("_") or name in dst_globals:
+            continue
+        dst_globals[name] = getattr(src_module, name)

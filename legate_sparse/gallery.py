@@ -1,0 +1,85 @@
+# SPDX-License-Identifier: Apache-2.0
+"""Matrix construction helpers: ``diags`` and ``eye``.
+
+Counterpart of the reference's ``legate_sparse/gallery.py`` (diags,
+gallery.py:77-195), scipy-compatible signature.
+"""
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from .dia import dia_array
+from .types import to_torch_dtype
+
+
+def diags(diagonals, offsets=0, shape=None, format=None, dtype=None):
+    """scipy.sparse.diags-compatible (reference gallery.py:77-195).
+
+    ``diagonals``: scalar / 1-D / list of scalars / list of 1-D arrays.
+    ``format``: "dia" (default) or "csr".
+    """
+    if format not in (None, "dia", "csr"):
+        raise NotImplementedError(f"format {format} not supported")
+    # normalize: list of (array, offset)
+    if np.isscalar(diagonals) or (
+            hasattr(diagonals, "ndim") and getattr(diagonals, "ndim") == 1
+            and not isinstance(diagonals, (list, tuple))):
+        diagonals = [np.atleast_1d(np.asarray(diagonals))]
+        offsets = np.atleast_1d(np.asarray(offsets))
+    else:
+        diagonals = [np.atleast_1d(np.asarray(d)) for d in diagonals]
+        offsets = np.atleast_1d(np.asarray(offsets))
+    if len(diagonals) != len(offsets):
+        raise ValueError("number of diagonals does not match offsets")
+
+    if dtype is None:
+        dtype = np.promote_types(
+            np.result_type(*[d.dtype for d in diagonals]), np.float32)
+        if dtype == np.float16:
+            dtype = np.float32
+    tdt = to_torch_dtype(dtype)
+
+    # infer shape (scipy rule: square matrix fitting all diagonals)
+    if shape is None:
+        M = max(len(d) + abs(int(o)) for d, o in zip(diagonals, offsets))
+        shape = (M, M)
+    M, N = int(shape[0]), int(shape[1])
+
+    n_diags = len(offsets)
+    data = torch.zeros(n_diags, N, dtype=tdt)
+    for k, (diag, off) in enumerate(zip(diagonals, offsets)):
+        off = int(off)
+        start = max(0, off)
+        end = min(N, M + off)
+        length = end - start
+        if length < 0:
+            raise ValueError(f"offset {off} out of bounds for shape {shape}")
+        d = torch.from_numpy(np.ascontiguousarray(diag)).to(tdt).reshape(-1)
+        if d.numel() == 1:
+            data[k, start:end] = d.item() if not d.is_complex() else d[0]
+        else:
+            if d.numel() < length:
+                raise ValueError(
+                    f"diagonal {k} (offset {off}) too short: "
+                    f"{d.numel()} < {length}")
+            data[k, start:end] = d[:length]
+    out = dia_array((data, torch.from_numpy(
+        np.ascontiguousarray(offsets, dtype=np.int64))), shape=(M, N))
+    if format == "csr":
+        return out.tocsr()
+    return out
+
+
+def eye(m, n=None, k=0, dtype=np.float64, format=None):
+    """scipy.sparse.eye-compatible identity/shifted-diagonal builder."""
+    if n is None:
+        n = m
+    m, n = int(m), int(n)
+    length = max(0, min(m + min(k, 0), n - max(k, 0)))
+    return diags([np.ones(length, dtype=dtype)], [k], shape=(m, n),
+                 format=format or "dia", dtype=dtype)
+
+
+def identity(n, dtype=np.float64, format=None):
+    return eye(n, dtype=dtype, format=format)

@@ -1,0 +1,100 @@
+# SPDX-License-Identifier: Apache-2.0
+"""MatrixMarket IO.
+
+Counterpart of the reference's ``legate_sparse/io.py`` (mmread, io.py:27-55)
+and its C++ single-task parser (src/sparse/io/mtx_to_coo.cc:31-143): header
+parse (type/field/symmetry), COO triples, symmetric duplication.  Here the
+parse is a vectorized numpy pass (every rank parses; the COO constructor
+keeps only local rows).
+"""
+from __future__ import annotations
+
+import numpy as np
+
+from .csr import csr_array
+from .types import coord_ty
+
+
+def mmread(source) -> csr_array:
+    """Read a MatrixMarket coordinate file into a (distributed) csr_array."""
+    close = False
+    if isinstance(source, (str, bytes)):
+        f = open(source, "rb")
+        close = True
+    else:
+        f = source
+    try:
+        header = f.readline().decode().strip().lower()
+        parts = header.split()
+        if len(parts) != 5 or parts[0] != "%%matrixmarket":
+            raise ValueError(f"bad MatrixMarket banner: {header}")
+        _, obj, fmt, field, symmetry = parts
+        if obj != "matrix" or fmt != "coordinate":
+            raise NotImplementedError(
+                "only 'matrix coordinate' files are supported")
+        if field not in ("real", "integer", "pattern", "complex"):
+            raise NotImplementedError(f"field {field} not supported")
+        # skip comments
+        line = f.readline().decode()
+        while line.startswith("%"):
+            line = f.readline().decode()
+        m, n, nnz = (int(tok) for tok in line.split())
+        body = f.read()
+    finally:
+        if close:
+            f.close()
+
+    import io as _io
+
+    ncols = {"pattern": 2, "real": 3, "integer": 3, "complex": 4}[field]
+    arr = (np.loadtxt(_io.BytesIO(body), ndmin=2) if nnz
+           else np.zeros((0, ncols)))
+    if arr.size and arr.shape[1] != ncols:
+        raise ValueError("malformed MatrixMarket body")
+    rows = arr[:, 0].astype(coord_ty) - 1
+    cols = arr[:, 1].astype(coord_ty) - 1
+    if field == "pattern":
+        vals = np.ones(rows.shape[0], dtype=np.float64)
+    elif field == "complex":
+        vals = arr[:, 2] + 1j * arr[:, 3]
+    else:
+        vals = arr[:, 2].astype(np.float64)
+
+    if symmetry in ("symmetric", "skew-symmetric", "hermitian"):
+        off = rows != cols
+        r2, c2, v2 = cols[off], rows[off], vals[off]
+        if symmetry == "skew-symmetric":
+            v2 = -v2
+        elif symmetry == "hermitian":
+            v2 = np.conj(v2)
+        rows = np.concatenate([rows, r2])
+        cols = np.concatenate([cols, c2])
+        vals = np.concatenate([vals, v2])
+    elif symmetry != "general":
+        raise NotImplementedError(f"symmetry {symmetry} not supported")
+
+    return csr_array((vals, (rows, cols)), shape=(m, n))
+
+
+def mmwrite(target, a) -> None:
+    """Minimal MatrixMarket writer (the reference has none; provided for
+    round-tripping).  Gathers nothing: rank 0 writes only at world_size==1."""
+    from .runtime import runtime
+
+    if runtime.world_size > 1:
+        raise NotImplementedError("mmwrite is single-process only")
+    A = a.tocsr() if hasattr(a, "tocsr") else csr_array(a)
+    grow, gcol, gval = A.tocoo()
+    rows = grow.cpu().numpy() + 1
+    cols = gcol.cpu().numpy() + 1
+    vals = gval.cpu().numpy()
+    field = "complex" if np.iscomplexobj(vals) else "real"
+    with open(target, "w") as f:
+        f.write(f"%%MatrixMarket matrix coordinate {field} general\n")
+        f.write(f"{A.shape[0]} {A.shape[1]} {rows.shape[0]}\n")
+        if field == "complex":
+            for r, c, v in zip(rows, cols, vals):
+                f.write(f"{r} {c} {v.real:.17g} {v.imag:.17g}\n")
+        else:
+            for r, c, v in zip(rows, cols, vals):
+                f.write(f"{r} {c} {v:.17g}\n")

@@ -1,0 +1,310 @@
+# SPDX-License-Identifier: Apache-2.0
+"""Iterative solvers: CG, GMRES, LinearOperator.
+
+Counterpart of the reference's ``legate_sparse/linalg.py`` (linalg.py:85-668).
+The key property preserved is the *async solver pipeline* (SURVEY §3.4):
+
+- ``cg_axpby`` consumes its scalars as 1-element DEVICE tensors; the
+  division a/b happens inside the fused HIP kernel (reference axpby.cu:25-47)
+  so no ``.item()`` ever occurs in the iteration body.
+- Scalar reductions (rho, pq) are local block-reduce + RCCL all-reduce of a
+  single element, left on device.
+- Convergence is tested only every ``conv_test_iters`` iterations
+  (reference linalg.py:529-533) — the only host syncs in the loop.
+
+All vectors are LOCAL shards of partition(n) in SPMD mode.
+"""
+from __future__ import annotations
+
+import math
+from typing import Callable, Optional
+
+import numpy as np
+import torch
+
+from . import comm, ops
+from .base import as_torch_1d
+from .runtime import runtime
+from .types import to_torch_dtype
+
+
+def _to_local_vec(v, n_global: int, dtype, device) -> torch.Tensor:
+    """Accept a global (replicated) or local-shard vector; return the
+    local shard."""
+    t = as_torch_1d(v, device=device).to(to_torch_dtype(dtype))
+    part = runtime.partition(n_global)
+    lo, hi = part.lo(runtime.rank), part.hi(runtime.rank)
+    if t.numel() == n_global and n_global != (hi - lo):
+        return t[lo:hi].contiguous()
+    if t.numel() == (hi - lo):
+        return t.contiguous()
+    if t.numel() == n_global:
+        return t.contiguous()
+    raise ValueError(
+        f"vector length {t.numel()} matches neither global {n_global} nor "
+        f"local shard {hi - lo}")
+
+
+def _gdot(x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    """Global <x, y> as a 1-element device tensor."""
+    s = ops.vdot(x, y, conj=True)
+    if runtime.world_size > 1:
+        comm.allreduce_(s)
+    return s
+
+
+def _gnorm(x: torch.Tensor) -> torch.Tensor:
+    s = _gdot(x, x)
+    return torch.sqrt(torch.abs(s))
+
+
+# ---------------------------------------------------------------------------
+# LinearOperator family (reference linalg.py:85-414)
+# ---------------------------------------------------------------------------
+class LinearOperator:
+    def __init__(self, shape, matvec=None, rmatvec=None, dtype=None):
+        if matvec is not None:
+            # scipy-style factory usage
+            self.shape = tuple(int(s) for s in shape)
+            self.dtype = np.dtype(dtype) if dtype is not None else None
+            self._matvec_fn = matvec
+            self._rmatvec_fn = rmatvec
+        else:
+            self.shape = tuple(int(s) for s in shape)
+            self.dtype = np.dtype(dtype) if dtype is not None else None
+            self._matvec_fn = None
+            self._rmatvec_fn = None
+
+    def matvec(self, x, out=None):
+        if self._matvec_fn is None:
+            raise NotImplementedError
+        y = self._matvec_fn(x)
+        if out is not None:
+            out.copy_(y if isinstance(y, torch.Tensor) else torch.as_tensor(y))
+            return out
+        return y
+
+    def rmatvec(self, x, out=None):
+        if self._rmatvec_fn is None:
+            raise NotImplementedError
+        y = self._rmatvec_fn(x)
+        if out is not None:
+            out.copy_(y if isinstance(y, torch.Tensor) else torch.as_tensor(y))
+            return out
+        return y
+
+    def __matmul__(self, x):
+        return self.matvec(x)
+
+    def dot(self, x):
+        return self.matvec(x)
+
+
+class IdentityOperator(LinearOperator):
+    def __init__(self, shape, dtype=None):
+        super().__init__(shape, dtype=dtype)
+
+    def matvec(self, x, out=None):
+        if out is not None:
+            out.copy_(x)
+            return out
+        return x.clone() if isinstance(x, torch.Tensor) else x
+
+    rmatvec = matvec
+
+
+class _SparseMatrixLinearOperator(LinearOperator):
+    """Wraps a csr_array; caches A.conj().T for rmatvec
+    (reference linalg.py:375-387)."""
+
+    def __init__(self, A):
+        self.A = A
+        self._AH = None
+        super().__init__(A.shape, dtype=A.dtype)
+
+    def matvec(self, x, out=None):
+        return self.A.dot(x, out=out)
+
+    def rmatvec(self, x, out=None):
+        if self._AH is None:
+            self._AH = self.A.conj().transpose()
+        return self._AH.dot(x, out=out)
+
+
+def aslinearoperator(A):
+    if isinstance(A, LinearOperator):
+        return A
+    if hasattr(A, "format"):
+        return _SparseMatrixLinearOperator(A if A.format == "csr"
+                                           else A.tocsr())
+    raise TypeError(f"cannot wrap {type(A)} as LinearOperator")
+
+
+# ---------------------------------------------------------------------------
+# Fused CG update (reference linalg.py:433-451 + axpby.cu)
+# ---------------------------------------------------------------------------
+def cg_axpby(y: torch.Tensor, x: torch.Tensor, a: torch.Tensor,
+             b: torch.Tensor, isalpha: bool, negate: bool = False):
+    """y = (±a/b)·x + y  (isalpha) or y = x + (±a/b)·y  — a, b are
+    1-element device tensors consumed inside the kernel."""
+    return ops.axpby(y, x, a, b, isalpha, negate)
+
+
+# ---------------------------------------------------------------------------
+# CG (reference linalg.py:465-535, CuPy-derived)
+# ---------------------------------------------------------------------------
+def cg(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
+       maxiter: Optional[int] = None, M=None,
+       callback: Optional[Callable] = None, conv_test_iters: int = 25):
+    """Conjugate gradient. Returns (x, info): info=0 converged, >0 not."""
+    Aop = aslinearoperator(A)
+    n = Aop.shape[0]
+    dtype = Aop.dtype if Aop.dtype is not None else np.float64
+    device = runtime.device
+    b = _to_local_vec(b, n, dtype, device)
+    if maxiter is None:
+        maxiter = n * 10
+    Mop = aslinearoperator(M) if M is not None and not isinstance(
+        M, LinearOperator) else (M or IdentityOperator(Aop.shape,
+                                                       dtype=dtype))
+
+    bnrm2 = _gnorm(b)
+    atol_t = torch.clamp(bnrm2 * rtol, min=atol)
+
+    if x0 is None:
+        x = torch.zeros_like(b)
+        r = b.clone()
+    else:
+        x = _to_local_vec(x0, n, dtype, device).clone()
+        r = b - Aop.matvec(x)
+
+    z = Mop.matvec(r)
+    if not isinstance(z, torch.Tensor):
+        z = torch.as_tensor(z, device=device)
+    if z.data_ptr() == r.data_ptr():
+        z = z.clone()
+    p = z.clone()
+    q = torch.empty_like(b)
+    rho = _gdot(r, z)
+    iters = 0
+    info = maxiter
+
+    while iters < maxiter:
+        Aop.matvec(p, out=q)
+        pq = _gdot(p, q)
+        # alpha = rho / pq ; x += alpha p ; r -= alpha q — all in-kernel
+        cg_axpby(x, p, rho, pq, isalpha=True, negate=False)
+        cg_axpby(r, q, rho, pq, isalpha=True, negate=True)
+        iters += 1
+        if callback is not None:
+            callback(x)
+        if conv_test_iters > 0 and (iters % conv_test_iters == 0
+                                    or iters == maxiter):
+            rnorm = _gnorm(r)  # host sync point (deliberate, amortized)
+            if bool((rnorm <= atol_t).item()):
+                info = 0
+                break
+        z = Mop.matvec(r, out=z if isinstance(z, torch.Tensor) else None)
+        if not isinstance(z, torch.Tensor):
+            z = torch.as_tensor(z, device=device)
+        rho1 = rho
+        rho = _gdot(r, z)
+        # beta = rho / rho1 ; p = z + beta p
+        cg_axpby(p, z, rho, rho1, isalpha=False, negate=False)
+
+    if info != 0:
+        rnorm = _gnorm(r)
+        if bool((rnorm <= atol_t).item()):
+            info = 0
+    return x, info
+
+
+# ---------------------------------------------------------------------------
+# GMRES (reference linalg.py:540-668, restarted, host lstsq)
+# ---------------------------------------------------------------------------
+def gmres(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
+          restart: Optional[int] = None, maxiter: Optional[int] = None,
+          M=None, callback: Optional[Callable] = None):
+    """Restarted GMRES.  Arnoldi basis vectors are local shards; the small
+    Hessenberg solve runs on host numpy (reference linalg.py:540-668)."""
+    Aop = aslinearoperator(A)
+    n = Aop.shape[0]
+    dtype = Aop.dtype if Aop.dtype is not None else np.float64
+    device = runtime.device
+    b = _to_local_vec(b, n, dtype, device)
+    if maxiter is None:
+        maxiter = min(n, 1000)
+    if restart is None:
+        restart = min(20, n)
+    restart = min(restart, maxiter)
+    Mop = M if isinstance(M, LinearOperator) else (
+        aslinearoperator(M) if M is not None
+        else IdentityOperator(Aop.shape, dtype=dtype))
+
+    bnrm2 = float(_gnorm(b).item())
+    tol = max(rtol * bnrm2, atol)
+    if bnrm2 == 0.0:
+        return b.clone(), 0
+
+    if x0 is None:
+        x = torch.zeros_like(b)
+    else:
+        x = _to_local_vec(x0, n, dtype, device).clone()
+
+    cdtype = to_torch_dtype(dtype)
+    iters = 0
+    info = 1
+    while iters < maxiter:
+        r = b - Aop.matvec(x)
+        r = Mop.matvec(r)
+        if not isinstance(r, torch.Tensor):
+            r = torch.as_tensor(r, device=device)
+        beta = float(_gnorm(r).item())
+        if beta <= tol:
+            info = 0
+            break
+        m = min(restart, maxiter - iters)
+        V = torch.zeros(m + 1, r.numel(), dtype=cdtype, device=device)
+        H = np.zeros((m + 1, m),
+                     dtype=np.complex128 if np.dtype(dtype).kind == "c"
+                     else np.float64)
+        V[0] = r / beta
+        j_done = m
+        for j in range(m):
+            w = Aop.matvec(V[j])
+            w = Mop.matvec(w)
+            if not isinstance(w, torch.Tensor):
+                w = torch.as_tensor(w, device=device)
+            # Gram-Schmidt (batched dot per column, allreduced)
+            basis = V[:j + 1].conj() if w.is_complex() else V[:j + 1]
+            hcol = basis @ w.reshape(-1, 1)
+            if runtime.world_size > 1:
+                comm.allreduce_(hcol)
+            w = w - (hcol.reshape(1, -1) @ V[:j + 1]).reshape(-1)
+            hnorm = float(_gnorm(w).item())
+            H[: j + 1, j] = hcol.reshape(-1).cpu().numpy()
+            H[j + 1, j] = hnorm
+            iters += 1
+            if hnorm <= 1e-14 * max(beta, 1.0):
+                j_done = j + 1
+                break
+            V[j + 1] = w / hnorm
+            j_done = j + 1
+        e1 = np.zeros(j_done + 1, dtype=H.dtype)
+        e1[0] = beta
+        ym, *_ = np.linalg.lstsq(H[: j_done + 1, : j_done], e1, rcond=None)
+        yt = torch.from_numpy(np.ascontiguousarray(ym)).to(
+            device=device, dtype=cdtype)
+        x = x + (yt.reshape(1, -1) @ V[:j_done]).reshape(-1)
+        if callback is not None:
+            callback(x)
+        res = float(_gnorm(b - Aop.matvec(x)).item())
+        if res <= tol:
+            info = 0
+            break
+    else:
+        pass
+    if info != 0:
+        res = float(_gnorm(b - Aop.matvec(x)).item())
+        info = 0 if res <= tol else maxiter
+    return x, info

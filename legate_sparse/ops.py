@@ -1,0 +1,258 @@
+# SPDX-License-Identifier: Apache-2.0
+"""Internal op dispatch: HIP kernels on GPU, C++/OpenMP or torch on CPU.
+
+Each op here corresponds to one task class of the reference's C++ layer
+(SURVEY §2.2/§2.3): CSRSpMVRowSplit → ``spmv``; SpGEMMCSRxCSRxCSR{NNZ,} →
+``spgemm_local``; CSRToDense/DenseToCSR → conversions; GetCSRDiagonal →
+``diagonal``; AXPBY → ``axpby``; plus the dot/norm block-reductions feeding
+RCCL all-reduce.
+
+GPU tensors REQUIRE the in-tree HIP extension — there is no silent eager
+fallback (set LS_FORCE_FALLBACK=1 explicitly to debug against torch).
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from . import _cext
+from .settings import settings
+
+_DTYPE_CODE = {
+    torch.float32: 0,
+    torch.float64: 1,
+    torch.complex64: 2,
+    torch.complex128: 3,
+}
+
+
+def _code(t: torch.Tensor) -> int:
+    return _DTYPE_CODE[t.dtype]
+
+
+def _use_hip(t: torch.Tensor) -> bool:
+    return t.is_cuda and not settings.force_cpu_fallback
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+# ---------------------------------------------------------------------------
+# SpMV: y[i] (+)= sum_j vals[jp] * x[indices[jp]]   (rows local, x gathered)
+# ---------------------------------------------------------------------------
+def spmv(indptr: torch.Tensor, indices: torch.Tensor, vals: torch.Tensor,
+         x: torch.Tensor, y: Optional[torch.Tensor] = None,
+         accumulate: bool = False) -> torch.Tensor:
+    n_rows = indptr.numel() - 1
+    if y is None:
+        y = torch.empty(n_rows, dtype=vals.dtype, device=vals.device)
+        accumulate = False
+    if _use_hip(vals):
+        ext = _cext.require_hip()
+        ext.spmv(indptr.data_ptr(), indices.data_ptr(), vals.data_ptr(),
+                 x.data_ptr(), y.data_ptr(), n_rows, vals.numel(),
+                 _code(vals), accumulate, _stream())
+        return y
+    if not vals.is_cuda and _cext.has_cpu():
+        _cext.require_cpu().spmv(indptr.data_ptr(), indices.data_ptr(),
+                                 vals.data_ptr(), x.data_ptr(), y.data_ptr(),
+                                 n_rows, _code(vals), accumulate)
+        return y
+    # torch fallback (debug / extension-less CPU)
+    prod = vals * x[indices]
+    row_ids = torch.repeat_interleave(
+        torch.arange(n_rows, device=vals.device),
+        (indptr[1:] - indptr[:-1]),
+    )
+    if not accumulate:
+        y.zero_()
+    y.index_add_(0, row_ids, prod)
+    return y
+
+
+# ---------------------------------------------------------------------------
+# SpGEMM (local rows of A) x (gathered rows of B) -> local rows of C
+# ---------------------------------------------------------------------------
+def spgemm_local(
+    A_indptr: torch.Tensor, A_indices: torch.Tensor, A_vals: torch.Tensor,
+    B_indptr: torch.Tensor, B_indices: torch.Tensor, B_vals: torch.Tensor,
+    n_colsB: int,
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Gustavson 2-phase; B rows indexable by A's (global) column ids."""
+    n_rows = A_indptr.numel() - 1
+    dev = A_vals.device
+    if _use_hip(A_vals):
+        ext = _cext.require_hip()
+        row_nnz = torch.empty(n_rows, dtype=torch.int64, device=dev)
+        ext.spgemm_symbolic(
+            A_indptr.data_ptr(), A_indices.data_ptr(), n_rows,
+            B_indptr.data_ptr(), B_indices.data_ptr(), int(n_colsB),
+            row_nnz.data_ptr(), _stream())
+        C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
+        nnz = int(C_indptr[-1].item())
+        C_indices = torch.empty(nnz, dtype=torch.int64, device=dev)
+        C_vals = torch.empty(nnz, dtype=A_vals.dtype, device=dev)
+        ext.spgemm_numeric(
+            A_indptr.data_ptr(), A_indices.data_ptr(), A_vals.data_ptr(),
+            n_rows, B_indptr.data_ptr(), B_indices.data_ptr(),
+            B_vals.data_ptr(), int(n_colsB), C_indptr.data_ptr(),
+            C_indices.data_ptr(), C_vals.data_ptr(), _code(A_vals),
+            _stream())
+        return C_indptr, C_indices, C_vals
+    if not A_vals.is_cuda and _cext.has_cpu():
+        ext = _cext.require_cpu()
+        row_nnz = torch.empty(n_rows, dtype=torch.int64)
+        ext.spgemm_symbolic(A_indptr.data_ptr(), A_indices.data_ptr(),
+                            n_rows, B_indptr.data_ptr(),
+                            B_indices.data_ptr(), int(n_colsB),
+                            row_nnz.data_ptr())
+        C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64)
+        torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
+        nnz = int(C_indptr[-1])
+        C_indices = torch.empty(nnz, dtype=torch.int64)
+        C_vals = torch.empty(nnz, dtype=A_vals.dtype)
+        ext.spgemm_numeric(A_indptr.data_ptr(), A_indices.data_ptr(),
+                           A_vals.data_ptr(), n_rows, B_indptr.data_ptr(),
+                           B_indices.data_ptr(), B_vals.data_ptr(),
+                           int(n_colsB), C_indptr.data_ptr(),
+                           C_indices.data_ptr(), C_vals.data_ptr(),
+                           _code(A_vals))
+        return C_indptr, C_indices, C_vals
+    # torch fallback: ESC (expand - sort - compress), works on any device.
+    return _spgemm_esc(A_indptr, A_indices, A_vals, B_indptr, B_indices,
+                       B_vals, n_colsB)
+
+
+def _spgemm_esc(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
+                n_colsB):
+    dev = A_vals.device
+    n_rows = A_indptr.numel() - 1
+    a_rows = torch.repeat_interleave(
+        torch.arange(n_rows, device=dev), A_indptr[1:] - A_indptr[:-1])
+    k = A_indices
+    blen = (B_indptr[1:] - B_indptr[:-1])[k]
+    e_rows = torch.repeat_interleave(a_rows, blen)
+    e_avals = torch.repeat_interleave(A_vals, blen)
+    total = int(blen.sum())
+    # position within each expanded B-row segment
+    seg_starts = torch.cumsum(blen, 0) - blen
+    pos = torch.arange(total, device=dev) - torch.repeat_interleave(
+        seg_starts, blen)
+    b_off = torch.repeat_interleave(B_indptr[k], blen) + pos
+    e_cols = B_indices[b_off]
+    e_vals = e_avals * B_vals[b_off]
+    key = e_rows * int(n_colsB) + e_cols
+    key_sorted, order = torch.sort(key, stable=True)
+    vals_sorted = e_vals[order]
+    uniq, inverse = torch.unique_consecutive(key_sorted, return_inverse=True)
+    C_vals = torch.zeros(uniq.numel(), dtype=A_vals.dtype, device=dev)
+    C_vals.index_add_(0, inverse, vals_sorted)
+    C_rows = torch.div(uniq, n_colsB, rounding_mode="floor")
+    C_indices = uniq - C_rows * n_colsB
+    row_nnz = torch.bincount(C_rows, minlength=n_rows)
+    C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
+    return C_indptr, C_indices, C_vals
+
+
+# ---------------------------------------------------------------------------
+# Conversions
+# ---------------------------------------------------------------------------
+def csr_to_dense(indptr, indices, vals, n_rows, n_cols) -> torch.Tensor:
+    out = torch.zeros(n_rows, n_cols, dtype=vals.dtype, device=vals.device)
+    if _use_hip(vals):
+        ext = _cext.require_hip()
+        ext.csr_to_dense(indptr.data_ptr(), indices.data_ptr(),
+                         vals.data_ptr(), out.data_ptr(), n_rows, n_cols,
+                         _code(vals), _stream())
+        return out
+    row_ids = torch.repeat_interleave(
+        torch.arange(n_rows, device=vals.device), indptr[1:] - indptr[:-1])
+    out[row_ids, indices] = vals
+    return out
+
+
+def dense_to_csr(dense: torch.Tensor):
+    n_rows, n_cols = dense.shape
+    dense = dense.contiguous()
+    if _use_hip(dense):
+        ext = _cext.require_hip()
+        row_nnz = torch.empty(n_rows, dtype=torch.int64, device=dense.device)
+        ext.dense_to_csr_nnz(dense.data_ptr(), row_nnz.data_ptr(), n_rows,
+                             n_cols, _code(dense), _stream())
+        indptr = torch.zeros(n_rows + 1, dtype=torch.int64,
+                             device=dense.device)
+        torch.cumsum(row_nnz, dim=0, out=indptr[1:])
+        nnz = int(indptr[-1].item())
+        indices = torch.empty(nnz, dtype=torch.int64, device=dense.device)
+        vals = torch.empty(nnz, dtype=dense.dtype, device=dense.device)
+        ext.dense_to_csr_fill(dense.data_ptr(), indptr.data_ptr(),
+                              indices.data_ptr(), vals.data_ptr(), n_rows,
+                              n_cols, _code(dense), _stream())
+        return indptr, indices, vals
+    mask = dense != 0
+    row_nnz = mask.sum(dim=1)
+    indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dense.device)
+    torch.cumsum(row_nnz, dim=0, out=indptr[1:])
+    nz = mask.nonzero(as_tuple=True)
+    return indptr, nz[1].to(torch.int64), dense[nz]
+
+
+def diagonal(indptr, indices, vals, n_rows, row_offset: int) -> torch.Tensor:
+    """diag[i] = vals[jp] where indices[jp] == row_offset + i, else 0
+    (reference get_diagonal.cu:26-44; row_offset makes it partition-aware)."""
+    out = torch.zeros(n_rows, dtype=vals.dtype, device=vals.device)
+    if _use_hip(vals):
+        ext = _cext.require_hip()
+        ext.diagonal(indptr.data_ptr(), indices.data_ptr(), vals.data_ptr(),
+                     out.data_ptr(), n_rows, int(row_offset), _code(vals),
+                     _stream())
+        return out
+    row_ids = torch.repeat_interleave(
+        torch.arange(n_rows, device=vals.device), indptr[1:] - indptr[:-1])
+    hit = indices == (row_ids + row_offset)
+    out[row_ids[hit]] = vals[hit]
+    return out
+
+
+# ---------------------------------------------------------------------------
+# Solver primitives
+# ---------------------------------------------------------------------------
+def axpby(y: torch.Tensor, x: torch.Tensor, a: torch.Tensor,
+          b: torch.Tensor, isalpha: bool, negate: bool) -> torch.Tensor:
+    """Fused CG update with DEVICE-scalar a, b (1-element tensors):
+        val = (negate ? -1 : 1) * a/b
+        isalpha: y = val*x + y      else: y = x + val*y
+    The division happens inside the kernel — no host sync (reference
+    axpby.cu:25-47, launched from linalg.py:433-451)."""
+    if _use_hip(y):
+        ext = _cext.require_hip()
+        ext.axpby(y.data_ptr(), x.data_ptr(), a.data_ptr(), b.data_ptr(),
+                  y.numel(), bool(isalpha), bool(negate), _code(y), _stream())
+        return y
+    val = a / b
+    if negate:
+        val = -val
+    if isalpha:
+        y.add_(x * val)
+    else:
+        y.mul_(val).add_(x)
+    return y
+
+
+def vdot(x: torch.Tensor, y: torch.Tensor,
+         conj: bool = True) -> torch.Tensor:
+    """Local <x, y> as a 1-element device tensor (block-reduce kernel on
+    GPU).  Caller all-reduces across ranks; never .item() in solver loops."""
+    if _use_hip(x):
+        ext = _cext.require_hip()
+        out = torch.zeros(1, dtype=x.dtype, device=x.device)
+        ext.vdot(x.data_ptr(), y.data_ptr(), out.data_ptr(), x.numel(),
+                 bool(conj), _code(x), _stream())
+        return out
+    if conj and x.is_complex():
+        return (x.conj() * y).sum().reshape(1)
+    return (x * y).sum().reshape(1)

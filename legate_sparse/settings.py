@@ -1,0 +1,43 @@
+# SPDX-License-Identifier: Apache-2.0
+"""Environment-variable configuration.
+
+Mirrors the reference's ``legate_sparse/settings.py`` (settings.py:22-48)
+with MI355X-native names:
+
+- ``LS_FAST_SPGEMM``    — analogue of LEGATE_SPARSE_FAST_SPGEMM (csr.py:674
+  in the reference selects cuSPARSE ALG1): selects the speed-over-memory
+  SpGEMM hash-table sizing on GPU.
+- ``LS_PRECISE_IMAGES`` — analogue of LEGATE_SPARSE_PRECISE_IMAGES: when
+  true, SpMV halo exchange ships exact per-rank column windows instead of
+  the full x all-gather.
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass
+
+
+def _env_bool(name: str, default: bool) -> bool:
+    v = os.environ.get(name)
+    if v is None:
+        return default
+    return v.strip().lower() in ("1", "true", "yes", "on")
+
+
+@dataclass
+class Settings:
+    @property
+    def fast_spgemm(self) -> bool:
+        return _env_bool("LS_FAST_SPGEMM", False)
+
+    @property
+    def precise_images(self) -> bool:
+        return _env_bool("LS_PRECISE_IMAGES", True)
+
+    @property
+    def force_cpu_fallback(self) -> bool:
+        """Testing hook: route GPU dispatch through the torch fallback."""
+        return _env_bool("LS_FORCE_FALLBACK", False)
+
+
+settings = Settings()

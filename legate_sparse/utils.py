@@ -1,0 +1,73 @@
+# SPDX-License-Identifier: Apache-2.0
+"""Utilities (reference legate_sparse/utils.py:28-202, re-designed for
+torch tensors instead of Legate stores)."""
+from __future__ import annotations
+
+import math
+from typing import Tuple
+
+import numpy as np
+import torch
+
+from . import coverage
+from .runtime import runtime
+from .types import (SUPPORTED_VALUE_DTYPES, is_supported_value_dtype,
+                    to_numpy_dtype, to_torch_dtype)
+
+SUPPORTED_DATATYPES = tuple(np.dtype(d) for d in SUPPORTED_VALUE_DTYPES)
+
+
+def is_dtype_supported(dt) -> bool:
+    return is_supported_value_dtype(dt)
+
+
+def find_common_type(a, b) -> np.dtype:
+    """Promote two array-likes' dtypes (reference utils.py:94-114)."""
+    da = a.dtype if hasattr(a, "dtype") else np.dtype(type(a))
+    db = b.dtype if hasattr(b, "dtype") else np.dtype(type(b))
+    ta, tb = to_torch_dtype(da), to_torch_dtype(db)
+    return to_numpy_dtype(torch.promote_types(ta, tb))
+
+
+def cast_to_common_type(a, b):
+    dt = find_common_type(a, b)
+    return a.astype(dt) if hasattr(a, "astype") else a, \
+        b.astype(dt) if hasattr(b, "astype") else b
+
+
+def factor_int(n: int) -> Tuple[int, int]:
+    """Near-square factorization (reference utils.py:118-124)."""
+    r = int(math.isqrt(n))
+    while n % r != 0:
+        r -= 1
+    return r, n // r
+
+
+def get_arr(x) -> np.ndarray:
+    """torch tensor / numpy → numpy (host) array."""
+    if isinstance(x, torch.Tensor):
+        return x.detach().cpu().numpy()
+    return np.asarray(x)
+
+
+def local_vector(global_array, n: int = None) -> torch.Tensor:
+    """Slice a replicated global 1-D array to this rank's shard."""
+    t = global_array if isinstance(global_array, torch.Tensor) else \
+        torch.from_numpy(np.ascontiguousarray(global_array))
+    n = n if n is not None else t.numel()
+    lo, hi = runtime.local_range(n)
+    return t[lo:hi].to(runtime.device).contiguous()
+
+
+def full_vector(local: torch.Tensor, n: int) -> torch.Tensor:
+    """All-gather a local shard back to the replicated global vector."""
+    from . import comm
+
+    part = runtime.partition(n)
+    return comm.allgatherv(local, part.counts())
+
+
+def enable_profiler_ranges(on: bool = True):
+    """Label public API calls with nvtx/roctx ranges so rocprof traces
+    attribute kernels to API calls (SURVEY §5 tracing)."""
+    coverage._PROFILE_RANGES = bool(on)
