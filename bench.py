@@ -1,0 +1,134 @@
+#!/usr/bin/env python3
+# SPDX-License-Identifier: Apache-2.0
+"""Flagship benchmark: SpMV GFLOP/s (+ SpGEMM nnz/s) on 5-pt Poisson CSR.
+
+Driver contract: ``python bench.py --gpus N --steps K --warmup W``
+(multi-GPU via torch.distributed.run, one rank per GPU over RCCL).
+One timed step = one distributed SpMV  y = A @ x  (fp64), including the
+x-halo all-gather.  Weak scaling: each GPU owns an nx*nx block of rows
+(global grid nx x (N*nx)).  A secondary timed section measures SpGEMM
+A @ A and reports nnz(C)/s.
+
+Metric/config match BASELINE.json: "SpMV GFLOP/s + SpGEMM nnz/s on 5-pt
+Poisson CSR", synthetic operator, fp64 (the reference's solvers run fp64).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch  # noqa: E402
+import torch.distributed as dist  # noqa: E402
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=100)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--nx", type=int, default=4096,
+                   help="per-GPU grid edge (rows per GPU = nx*nx)")
+    p.add_argument("--spgemm-iters", type=int, default=5)
+    args = p.parse_args()
+
+    import legate_sparse as lsp
+    from legate_sparse.gallery import poisson_2d
+    from legate_sparse.runtime import runtime
+
+    rank = runtime.rank
+    ws = runtime.world_size
+    n_gpus = max(ws, 1)
+    on_gpu = torch.cuda.is_available()
+    nx = args.nx
+    ny = nx * n_gpus  # weak scaling: fixed nx*nx rows per GPU
+
+    A = poisson_2d(nx, ny)
+    M = A.shape[0]
+    nnz = A.nnz
+    lo, hi = A.row_range
+    x = torch.rand(hi - lo, dtype=torch.float64, device=runtime.device)
+    y = torch.empty_like(x)
+
+    def sync():
+        if ws > 1:
+            dist.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    # ---- SpMV ----
+    for _ in range(args.warmup):
+        lsp.spmv(A, x, y)
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        lsp.spmv(A, x, y)
+    sync()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+
+    # MAX over ranks of elapsed time
+    if ws > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=runtime.device if on_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1e3
+    gflops = 2.0 * nnz * args.steps / elapsed / 1e9
+
+    # ---- SpGEMM (secondary metric) ----
+    sg_iters = max(args.spgemm_iters, 1)
+    C = A @ A  # warmup + nnz discovery
+    c_nnz = C.nnz
+    del C
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(sg_iters):
+        C = A @ A
+        del C
+    sync()
+    t1 = time.perf_counter()
+    sg_elapsed = t1 - t0
+    if ws > 1:
+        t = torch.tensor([sg_elapsed], dtype=torch.float64,
+                         device=runtime.device if on_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        sg_elapsed = float(t.item())
+    spgemm_nnz_per_s = c_nnz * sg_iters / sg_elapsed
+
+    if rank == 0:
+        out = {
+            "metric": "SpMV GFLOP/s on 5-pt Poisson CSR (fp64)",
+            "value": round(gflops, 3),
+            "unit": "GFLOP/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 6),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp64",
+            "data": "synthetic 5-pt Poisson operator, random x",
+            "config": {
+                "model": "5-pt 2-D Poisson CSR",
+                "grid": f"{nx}x{ny}",
+                "rows": M,
+                "nnz": nnz,
+                "parallelism": f"row-split dp{n_gpus}, x allgather halo",
+                "spgemm_nnz_per_s": round(spgemm_nnz_per_s, 1),
+                "spgemm_ms_per_iter": round(sg_elapsed / sg_iters * 1e3, 3),
+                "spgemm_out_nnz": c_nnz,
+                "device": "cuda" if on_gpu else "cpu",
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

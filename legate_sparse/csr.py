@@ -320,6 +320,10 @@ class csr_array(CompressedBase):
         if t.ndim != 1:
             raise NotImplementedError(
                 "only matrix-vector and matrix-matrix products supported")
+        if not (t.dtype.is_floating_point or t.dtype.is_complex):
+            raise NotImplementedError(
+                f"unsupported operand dtype {t.dtype} "
+                "(float32/float64/complex64/complex128 only)")
         dt = _common_dtype(self._data, t)
         x = self._coerce_vec(t).to(dt)
         A = self if self._data.dtype == dt else self.astype(

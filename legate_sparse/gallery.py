@@ -83,3 +83,36 @@ def eye(m, n=None, k=0, dtype=np.float64, format=None):
 
 def identity(n, dtype=np.float64, format=None):
     return eye(n, dtype=dtype, format=format)
+
+
+def poisson_2d(nx: int, ny: int, dtype=np.float64):
+    """5-point 2-D Poisson (Dirichlet) operator on an nx*ny grid, built
+    DIRECTLY as the distributed CSR (each rank constructs only its local
+    row block — no replicated intermediate).  This is the benchmark
+    operator of BASELINE.md (reference examples/pde.py builds it via
+    diags; diags works too, this is the scalable path)."""
+    import torch
+    from .csr import csr_array
+    from .runtime import runtime
+
+    tdt = to_torch_dtype(dtype)
+    M = nx * ny
+    lo, hi = runtime.local_range(M)
+    dev = runtime.device
+    r = torch.arange(lo, hi, device=dev, dtype=torch.int64)
+    j = r % nx
+    lr = hi - lo
+    cols = torch.stack([r - nx, r - 1, r, r + 1, r + nx], dim=1)
+    mask = torch.stack([
+        r >= nx, j > 0, torch.ones(lr, dtype=torch.bool, device=dev),
+        j < nx - 1, r < M - nx,
+    ], dim=1)
+    vals = torch.tensor([-1.0, -1.0, 4.0, -1.0, -1.0], dtype=tdt,
+                        device=dev).reshape(1, 5).expand(lr, 5)
+    indices = cols[mask]
+    data = vals[mask].contiguous()
+    counts = mask.sum(dim=1)
+    indptr = torch.zeros(lr + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(counts, dim=0, out=indptr[1:])
+    return csr_array(None, shape=(M, M),
+                     _local=(indptr, indices, data))

@@ -84,24 +84,8 @@ def spgemm_local(
     n_rows = A_indptr.numel() - 1
     dev = A_vals.device
     if _use_hip(A_vals):
-        ext = _cext.require_hip()
-        row_nnz = torch.empty(n_rows, dtype=torch.int64, device=dev)
-        ext.spgemm_symbolic(
-            A_indptr.data_ptr(), A_indices.data_ptr(), n_rows,
-            B_indptr.data_ptr(), B_indices.data_ptr(), int(n_colsB),
-            row_nnz.data_ptr(), _stream())
-        C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
-        torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
-        nnz = int(C_indptr[-1].item())
-        C_indices = torch.empty(nnz, dtype=torch.int64, device=dev)
-        C_vals = torch.empty(nnz, dtype=A_vals.dtype, device=dev)
-        ext.spgemm_numeric(
-            A_indptr.data_ptr(), A_indices.data_ptr(), A_vals.data_ptr(),
-            n_rows, B_indptr.data_ptr(), B_indices.data_ptr(),
-            B_vals.data_ptr(), int(n_colsB), C_indptr.data_ptr(),
-            C_indices.data_ptr(), C_vals.data_ptr(), _code(A_vals),
-            _stream())
-        return C_indptr, C_indices, C_vals
+        return _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices,
+                           B_vals, n_colsB)
     if not A_vals.is_cuda and _cext.has_cpu():
         ext = _cext.require_cpu()
         row_nnz = torch.empty(n_rows, dtype=torch.int64)
@@ -124,6 +108,88 @@ def spgemm_local(
     # torch fallback: ESC (expand - sort - compress), works on any device.
     return _spgemm_esc(A_indptr, A_indices, A_vals, B_indptr, B_indices,
                        B_vals, n_colsB)
+
+
+def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
+                n_colsB):
+    """Binned hash-table Gustavson on gfx950 (src/hip/spgemm.hip).
+
+    Rows are binned by their expansion upper bound so the per-row hash
+    table fits LDS; oversize rows fall back to an HBM workspace table
+    (the automatic analogue of the reference's cuSPARSE ALG1/ALG3 switch,
+    spgemm_csr_csr_csr.cu:196-216)."""
+    ext = _cext.require_hip()
+    dev = A_vals.device
+    st = _stream()
+    n_rows = A_indptr.numel() - 1
+    code = _code(A_vals)
+    if n_colsB >= 2 ** 32:
+        raise NotImplementedError("SpGEMM requires B.shape[1] < 2^32")
+
+    ub = torch.empty(n_rows, dtype=torch.int64, device=dev)
+    ext.spgemm_row_ub(A_indptr.data_ptr(), A_indices.data_ptr(),
+                      B_indptr.data_ptr(), ub.data_ptr(), n_rows, st)
+    bins = [
+        torch.nonzero(ub <= 128).reshape(-1),
+        torch.nonzero((ub > 128) & (ub <= 1024)).reshape(-1),
+        torch.nonzero((ub > 1024) & (ub <= 4096)).reshape(-1),
+    ]
+    rows_g = torch.nonzero(ub > 4096).reshape(-1)
+
+    row_nnz = torch.zeros(n_rows, dtype=torch.int64, device=dev)
+    for cfg, rows in enumerate(bins):
+        if rows.numel():
+            ext.spgemm_symbolic_lds(
+                cfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
+                A_indices.data_ptr(), B_indptr.data_ptr(),
+                B_indices.data_ptr(), row_nnz.data_ptr(), st)
+    g_keys = g_off = g_sizes = None
+    if rows_g.numel():
+        ub_g = ub[rows_g]
+        g_sizes = torch.pow(
+            2.0, torch.ceil(torch.log2(2.0 * ub_g.double()))).to(torch.int64)
+        g_off = torch.zeros_like(g_sizes)
+        if g_sizes.numel() > 1:
+            g_off[1:] = torch.cumsum(g_sizes[:-1], dim=0)
+        total = int((g_off[-1] + g_sizes[-1]).item())
+        g_keys = torch.full((total,), -1, dtype=torch.int32, device=dev)
+        ext.spgemm_symbolic_global(
+            rows_g.data_ptr(), rows_g.numel(), A_indptr.data_ptr(),
+            A_indices.data_ptr(), B_indptr.data_ptr(),
+            B_indices.data_ptr(), g_keys.data_ptr(), g_off.data_ptr(),
+            g_sizes.data_ptr(), row_nnz.data_ptr(), st)
+
+    C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
+    nnz = int(C_indptr[-1].item())
+    C_indices = torch.empty(nnz, dtype=torch.int64, device=dev)
+    C_vals = torch.empty(nnz, dtype=A_vals.dtype, device=dev)
+
+    for cfg, rows in enumerate(bins):
+        if rows.numel():
+            ext.spgemm_numeric_lds(
+                cfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
+                A_indices.data_ptr(), A_vals.data_ptr(),
+                B_indptr.data_ptr(), B_indices.data_ptr(),
+                B_vals.data_ptr(), C_indptr.data_ptr(),
+                C_indices.data_ptr(), C_vals.data_ptr(), code, st)
+    if rows_g.numel():
+        g_keys.fill_(-1)
+        g_vals = torch.zeros(g_keys.numel(), dtype=A_vals.dtype, device=dev)
+        ext.spgemm_numeric_global(
+            rows_g.data_ptr(), rows_g.numel(), A_indptr.data_ptr(),
+            A_indices.data_ptr(), A_vals.data_ptr(), B_indptr.data_ptr(),
+            B_indices.data_ptr(), B_vals.data_ptr(), g_keys.data_ptr(),
+            g_vals.data_ptr(), g_off.data_ptr(), g_sizes.data_ptr(),
+            C_indptr.data_ptr(), C_indices.data_ptr(), C_vals.data_ptr(),
+            code, st)
+
+    # canonical per-row column sort (hash compaction emits unsorted)
+    row_ids = torch.repeat_interleave(
+        torch.arange(n_rows, device=dev), C_indptr[1:] - C_indptr[:-1])
+    key = row_ids * int(n_colsB) + C_indices
+    order = torch.argsort(key)
+    return C_indptr, C_indices[order], C_vals[order]
 
 
 def _spgemm_esc(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,

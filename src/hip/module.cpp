@@ -1,0 +1,57 @@
+// SPDX-License-Identifier: Apache-2.0
+// pybind11 bindings for the gfx950 HIP kernels.
+//
+// Deliberately ATen-free: kernels take raw device pointers (data_ptr()),
+// sizes and the torch HIP stream handle — the .so links only HIP runtime +
+// Python, so it builds with bare hipcc and travels in-tree.
+
+#include <pybind11/pybind11.h>
+#include <cstdint>
+
+namespace py = pybind11;
+using i64 = int64_t;
+
+void ls_spmv(uintptr_t, uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64,
+             i64, int, bool, uintptr_t);
+void ls_spgemm_row_ub(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64,
+                      uintptr_t);
+void ls_spgemm_symbolic_lds(int, uintptr_t, i64, uintptr_t, uintptr_t,
+                            uintptr_t, uintptr_t, uintptr_t, uintptr_t);
+void ls_spgemm_numeric_lds(int, uintptr_t, i64, uintptr_t, uintptr_t,
+                           uintptr_t, uintptr_t, uintptr_t, uintptr_t,
+                           uintptr_t, uintptr_t, uintptr_t, int, uintptr_t);
+void ls_spgemm_symbolic_global(uintptr_t, i64, uintptr_t, uintptr_t,
+                               uintptr_t, uintptr_t, uintptr_t, uintptr_t,
+                               uintptr_t, uintptr_t, uintptr_t);
+void ls_spgemm_numeric_global(uintptr_t, i64, uintptr_t, uintptr_t,
+                              uintptr_t, uintptr_t, uintptr_t, uintptr_t,
+                              uintptr_t, uintptr_t, uintptr_t, uintptr_t,
+                              uintptr_t, uintptr_t, uintptr_t, int,
+                              uintptr_t);
+void ls_csr_to_dense(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, i64,
+                     int, uintptr_t);
+void ls_dense_to_csr_nnz(uintptr_t, uintptr_t, i64, i64, int, uintptr_t);
+void ls_dense_to_csr_fill(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64,
+                          i64, int, uintptr_t);
+void ls_diagonal(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, i64, int,
+                 uintptr_t);
+void ls_axpby(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, bool, bool,
+              int, uintptr_t);
+void ls_vdot(uintptr_t, uintptr_t, uintptr_t, i64, bool, int, uintptr_t);
+
+PYBIND11_MODULE(_hip_kernels, m) {
+  m.doc() = "legate_sparse gfx950 HIP kernels";
+  m.def("spmv", &ls_spmv);
+  m.def("spgemm_row_ub", &ls_spgemm_row_ub);
+  m.def("spgemm_symbolic_lds", &ls_spgemm_symbolic_lds);
+  m.def("spgemm_numeric_lds", &ls_spgemm_numeric_lds);
+  m.def("spgemm_symbolic_global", &ls_spgemm_symbolic_global);
+  m.def("spgemm_numeric_global", &ls_spgemm_numeric_global);
+  m.def("csr_to_dense", &ls_csr_to_dense);
+  m.def("dense_to_csr_nnz", &ls_dense_to_csr_nnz);
+  m.def("dense_to_csr_fill", &ls_dense_to_csr_fill);
+  m.def("diagonal", &ls_diagonal);
+  m.def("axpby", &ls_axpby);
+  m.def("vdot", &ls_vdot);
+  m.attr("arch") = "gfx950";
+}

@@ -1,0 +1,120 @@
+// SPDX-License-Identifier: Apache-2.0
+// Solver primitives for gfx950: fused AXPBY with device scalars, and
+// block-reduced dot products.
+//
+// AXPBY (reference axpby.cu:25-47): the scalar quotient a/b is computed
+// INSIDE the kernel from 1-element device tensors, so the CG loop never
+// synchronizes with the host.  Memory-bound: vectorized grid-stride.
+//
+// VDOT: grid-stride partial sums -> wave shuffle reduce -> LDS across
+// waves -> one device-scope atomicAdd per block into the 1-element output
+// (Guideline 12).  The output feeds a 1-element RCCL all-reduce.
+
+#include "common.h"
+
+namespace {
+
+template <typename T, bool IS_ALPHA, bool NEGATE>
+__global__ __launch_bounds__(LS_THREADS) void axpby_kernel(
+    T* __restrict__ y, const T* __restrict__ x, const T* __restrict__ a,
+    const T* __restrict__ b, i64 n) {
+  T val = (*a) / (*b);
+  if constexpr (NEGATE) {
+    if constexpr (is_cplx<T>::value)
+      val = ls_zero<T>() - val;
+    else
+      val = -val;
+  }
+  const i64 stride = (i64)gridDim.x * blockDim.x;
+  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if constexpr (IS_ALPHA)
+      y[i] = val * x[i] + y[i];
+    else
+      y[i] = x[i] + val * y[i];
+  }
+}
+
+__device__ inline void atomic_add_out(float* p, float v) { atomicAdd(p, v); }
+__device__ inline void atomic_add_out(double* p, double v) {
+  atomicAdd(p, v);
+}
+template <typename T>
+__device__ inline void atomic_add_out(Cplx<T>* p, Cplx<T> v) {
+  atomicAdd(&p->re, v.re);
+  atomicAdd(&p->im, v.im);
+}
+
+template <typename T, bool CONJ>
+__global__ __launch_bounds__(LS_THREADS) void vdot_kernel(
+    const T* __restrict__ x, const T* __restrict__ y, T* __restrict__ out,
+    i64 n) {
+  __shared__ T partials[LS_THREADS / WAVE_SIZE];
+  T acc = ls_zero<T>();
+  const i64 stride = (i64)gridDim.x * blockDim.x;
+  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if constexpr (CONJ)
+      acc += ls_conj(x[i]) * y[i];
+    else
+      acc += x[i] * y[i];
+  }
+  acc = group_reduce_sum<T, WAVE_SIZE>(acc);
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  if (lane == 0) partials[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    T total = partials[0];
+    for (int w = 1; w < LS_THREADS / WAVE_SIZE; ++w) total += partials[w];
+    atomic_add_out(out, total);
+  }
+}
+
+}  // namespace
+
+void ls_axpby(uintptr_t y, uintptr_t x, uintptr_t a, uintptr_t b, i64 n,
+              bool isalpha, bool negate, int dtype, uintptr_t stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  int grid = grid_1d(n, LS_THREADS, 4096);
+  DISPATCH_VAL_T(dtype, ({
+    auto yp = reinterpret_cast<val_t*>(y);
+    auto xp = reinterpret_cast<const val_t*>(x);
+    auto ap = reinterpret_cast<const val_t*>(a);
+    auto bp = reinterpret_cast<const val_t*>(b);
+    if (isalpha) {
+      if (negate)
+        hipLaunchKernelGGL((axpby_kernel<val_t, true, true>), dim3(grid),
+                           dim3(LS_THREADS), 0, s, yp, xp, ap, bp, n);
+      else
+        hipLaunchKernelGGL((axpby_kernel<val_t, true, false>), dim3(grid),
+                           dim3(LS_THREADS), 0, s, yp, xp, ap, bp, n);
+    } else {
+      if (negate)
+        hipLaunchKernelGGL((axpby_kernel<val_t, false, true>), dim3(grid),
+                           dim3(LS_THREADS), 0, s, yp, xp, ap, bp, n);
+      else
+        hipLaunchKernelGGL((axpby_kernel<val_t, false, false>), dim3(grid),
+                           dim3(LS_THREADS), 0, s, yp, xp, ap, bp, n);
+    }
+  }));
+  ls_check(hipGetLastError(), "axpby");
+}
+
+void ls_vdot(uintptr_t x, uintptr_t y, uintptr_t out, i64 n, bool conj,
+             int dtype, uintptr_t stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  int grid = grid_1d(n, LS_THREADS * 8, 2048);
+  DISPATCH_VAL_T(dtype, ({
+    auto xp = reinterpret_cast<const val_t*>(x);
+    auto yp = reinterpret_cast<const val_t*>(y);
+    auto op = reinterpret_cast<val_t*>(out);
+    if (conj)
+      hipLaunchKernelGGL((vdot_kernel<val_t, true>), dim3(grid),
+                         dim3(LS_THREADS), 0, s, xp, yp, op, n);
+    else
+      hipLaunchKernelGGL((vdot_kernel<val_t, false>), dim3(grid),
+                         dim3(LS_THREADS), 0, s, xp, yp, op, n);
+  }));
+  ls_check(hipGetLastError(), "vdot");
+}

@@ -1,0 +1,116 @@
+# SPDX-License-Identifier: Apache-2.0
+"""Worker for multi-process (gloo / RCCL) distributed checks.
+
+Launched by test_distributed.py via torch.distributed.run with
+world_size >= 2.  Every rank builds the same global inputs, the library
+shards them; results are compared against replicated scipy references.
+Exits nonzero on any mismatch.
+"""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import numpy as np
+import scipy.sparse as sp
+import torch
+import torch.distributed as dist
+
+import legate_sparse as lsp
+from legate_sparse.runtime import runtime
+from legate_sparse import utils as lsu
+from testutils import sample_csr, spd_system, banded_matrix, to_np
+
+
+def check(name, cond):
+    if not cond:
+        print(f"[rank {runtime.rank}] FAIL: {name}", flush=True)
+        sys.exit(1)
+    if runtime.rank == 0:
+        print(f"[rank 0] ok: {name}", flush=True)
+
+
+def main():
+    ws = runtime.world_size
+    assert ws >= 2, "needs torchrun world_size >= 2"
+    n = 103  # deliberately not divisible by world size
+
+    # --- SpMV ---
+    S = sample_csr(n, n, 0.2, seed=1)
+    A = lsp.csr_array(S)
+    x_g = np.random.default_rng(2).random(n)
+    y_local = A @ x_g  # replicated global x accepted
+    y_full = to_np(lsu.full_vector(y_local, n))
+    check("spmv", np.allclose(y_full, S @ x_g, rtol=1e-12))
+
+    # local-shard input path
+    x_local = lsu.local_vector(x_g, n)
+    y2 = A @ x_local
+    check("spmv local-shard input",
+          np.allclose(to_np(lsu.full_vector(y2, n)), S @ x_g, rtol=1e-12))
+
+    # --- SpGEMM ---
+    B_s = sample_csr(n, n, 0.2, seed=3)
+    C = A @ lsp.csr_array(B_s)
+    got = np.zeros((n, n))
+    lo, hi = C.row_range
+    got[lo:hi] = to_np(C.todense())
+    t = torch.from_numpy(got)
+    dist.all_reduce(t)
+    check("spgemm", np.allclose(t.numpy(), (S @ B_s).toarray(),
+                                rtol=1e-12, atol=1e-12))
+
+    # --- transpose ---
+    R_s = sample_csr(n, 57, 0.2, seed=4)
+    Rt = lsp.csr_array(R_s).T
+    got = np.zeros((57, n))
+    lo, hi = Rt.row_range
+    got[lo:hi] = to_np(Rt.todense())
+    t = torch.from_numpy(got)
+    dist.all_reduce(t)
+    check("transpose", np.allclose(t.numpy(), R_s.T.toarray()))
+
+    # --- diagonal / sum ---
+    Sq = sample_csr(n, n, 0.3, seed=5)
+    Aq = lsp.csr_array(Sq)
+    d_full = to_np(lsu.full_vector(Aq.diagonal(), n))
+    check("diagonal", np.allclose(d_full, Sq.diagonal()))
+    check("sum", abs(float(Aq.sum()) - Sq.sum()) < 1e-10)
+    check("nnz", Aq.nnz == Sq.nnz)
+
+    # --- CG ---
+    P = spd_system(96, seed=6)
+    b = np.random.default_rng(7).random(96)
+    x, info = lsp.linalg.cg(lsp.csr_array(P), b, rtol=1e-10, maxiter=600)
+    xf = to_np(lsu.full_vector(x, 96))
+    check("cg", info == 0 and
+          np.linalg.norm(P @ xf - b) <= 1e-8 * np.linalg.norm(b))
+
+    # --- GMRES ---
+    G = (sample_csr(80, 80, 0.25, seed=8) + 80 * sp.identity(80)).tocsr()
+    bg = np.random.default_rng(9).random(80)
+    xg, infog = lsp.linalg.gmres(lsp.csr_array(G), bg, rtol=1e-10,
+                                 restart=25, maxiter=400)
+    xgf = to_np(lsu.full_vector(xg, 80))
+    check("gmres", infog == 0 and
+          np.linalg.norm(G @ xgf - bg) <= 1e-8 * np.linalg.norm(bg))
+
+    # --- diags construction in SPMD ---
+    got = lsp.diags([np.ones(n), np.full(n - 1, 2.0)], [0, 1],
+                    shape=(n, n), format="csr")
+    want = sp.diags([np.ones(n), np.full(n - 1, 2.0)], [0, 1],
+                    shape=(n, n), format="csr")
+    lo, hi = got.row_range
+    dd = np.zeros((n, n))
+    dd[lo:hi] = to_np(got.todense())
+    t = torch.from_numpy(dd)
+    dist.all_reduce(t)
+    check("diags", np.allclose(t.numpy(), want.toarray()))
+
+    if runtime.rank == 0:
+        print("DIST_ALL_OK", flush=True)
+
+
+if __name__ == "__main__":
+    main()

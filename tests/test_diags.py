@@ -1,0 +1,80 @@
+# SPDX-License-Identifier: Apache-2.0
+"""diags/eye vs scipy (reference test_diags.py semantics)."""
+import numpy as np
+import pytest
+import scipy.sparse as sp
+
+import legate_sparse as lsp
+from testutils import to_np
+
+
+@pytest.mark.parametrize("offsets", [[0], [0, 1], [-1, 0, 1], [-2, 0, 3]])
+def test_diags_square(offsets):
+    n = 9
+    diagonals = [np.random.default_rng(o + 5).random(n - abs(o))
+                 for o in offsets]
+    got = lsp.diags(diagonals, offsets, shape=(n, n), format="csr")
+    want = sp.diags(diagonals, offsets, shape=(n, n), format="csr")
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray())
+
+
+def test_diags_rect():
+    got = lsp.diags([np.ones(4), np.full(4, 2.0)], [0, -1], shape=(5, 4),
+                    format="csr")
+    want = sp.diags([np.ones(4), np.full(4, 2.0)], [0, -1], shape=(5, 4),
+                    format="csr")
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray())
+
+
+def test_diags_scalar_broadcast():
+    got = lsp.diags([2.0], [0], shape=(6, 6), format="csr")
+    want = sp.diags([2.0], [0], shape=(6, 6), format="csr")
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray())
+
+
+def test_diags_infer_shape():
+    d = [np.arange(1.0, 5.0), np.arange(1.0, 4.0)]
+    got = lsp.diags(d, [0, 1], format="csr")
+    want = sp.diags(d, [0, 1], format="csr")
+    assert got.shape == want.shape
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray())
+
+
+def test_diags_dia_format():
+    got = lsp.diags([np.ones(5)], [0])
+    assert got.format == "dia"
+    np.testing.assert_allclose(to_np(got.tocsr().todense()), np.eye(5))
+
+
+def test_eye_identity():
+    np.testing.assert_allclose(
+        to_np(lsp.eye(7, format="csr").todense()), np.eye(7))
+    np.testing.assert_allclose(
+        to_np(lsp.eye(5, 8, k=2, format="csr").todense()),
+        sp.eye(5, 8, k=2).toarray())
+    np.testing.assert_allclose(
+        to_np(lsp.identity(4, format="csr").todense()), np.eye(4))
+
+
+def test_dia_array_tocsr():
+    data = np.array([[1.0, 2, 3, 4], [5, 6, 7, 8]])
+    offsets = np.array([0, -1])
+    got = lsp.dia_array((data, offsets), shape=(4, 4)).tocsr()
+    want = sp.dia_array((data, offsets), shape=(4, 4)).tocsr()
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray())
+
+
+def test_dia_transpose():
+    data = np.array([[1.0, 2, 3, 4], [5, 6, 7, 8]])
+    offsets = np.array([0, 1])
+    got = lsp.dia_array((data, offsets), shape=(4, 4)).T.tocsr()
+    want = sp.dia_array((data, offsets), shape=(4, 4)).T.tocsr()
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray())
+
+
+def test_dia_nnz():
+    data = np.ones((2, 5))
+    offsets = np.array([0, 2])
+    got = lsp.dia_array((data, offsets), shape=(5, 5))
+    want = sp.dia_array((data, offsets), shape=(5, 5))
+    assert got.nnz == want.nnz

@@ -1,0 +1,221 @@
+# SPDX-License-Identifier: Apache-2.0
+"""GPU numerics: every HIP kernel vs a plain fp32/fp64 torch/scipy
+reference (run on the MI355X box via gpurun; pytest -m gpu)."""
+import numpy as np
+import pytest
+import scipy.sparse as sp
+
+import legate_sparse as lsp
+from testutils import sample_csr, banded_matrix, to_np
+
+torch = pytest.importorskip("torch")
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="no GPU")
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _check_native():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from legate_sparse import _cext
+    # GPU runs MUST use the gfx950 extension — no silent fallback
+    _cext.require_hip()
+    yield
+
+
+@pytest.mark.parametrize("dtype",
+                         [np.float32, np.float64, np.complex64,
+                          np.complex128])
+def test_gpu_spmv_dtypes(dtype):
+    S = sample_csr(500, 431, 0.05, seed=1, dtype=dtype)
+    rng = np.random.default_rng(2)
+    x = rng.standard_normal(431).astype(dtype)
+    if np.dtype(dtype).kind == "c":
+        x = x + 1j * rng.standard_normal(431).astype(dtype)
+    A = lsp.csr_array(S)
+    assert A.data.is_cuda
+    y = A @ torch.from_numpy(x).cuda()
+    tol = 1e-4 if np.dtype(dtype).itemsize <= 8 else 1e-12
+    np.testing.assert_allclose(to_np(y), S @ x, rtol=tol, atol=tol)
+
+
+@pytest.mark.parametrize("nnz_per_row", [1, 3, 5, 11, 33, 101])
+def test_gpu_spmv_row_lengths(nnz_per_row):
+    # sweeps the W (sub-wave width) selection paths of spmv.hip
+    n = 2000
+    S = banded_matrix(n, nnz_per_row=min(nnz_per_row, n // 2) | 1)
+    x = np.random.default_rng(3).standard_normal(n)
+    y = lsp.csr_array(S) @ torch.from_numpy(x).cuda()
+    np.testing.assert_allclose(to_np(y), S @ x, rtol=1e-12)
+
+
+def test_gpu_spmv_long_single_row():
+    # one dense row among short rows (wave-loop path)
+    n = 1000
+    D = np.zeros((n, n))
+    D[0, :] = np.random.default_rng(4).standard_normal(n)
+    D[np.arange(1, n), np.arange(1, n)] = 2.0
+    S = sp.csr_array(D)
+    x = np.random.default_rng(5).standard_normal(n)
+    y = lsp.csr_array(S) @ torch.from_numpy(x).cuda()
+    np.testing.assert_allclose(to_np(y), S @ x, rtol=1e-12)
+
+
+@pytest.mark.parametrize("dtype", [np.float32, np.float64, np.complex128])
+def test_gpu_spgemm_small_bin(dtype):
+    A = banded_matrix(300, nnz_per_row=5).astype(dtype)
+    C = lsp.csr_array(A) @ lsp.csr_array(A)
+    tol = 1e-4 if np.dtype(dtype).itemsize <= 8 else 1e-12
+    np.testing.assert_allclose(to_np(C.todense()), (A @ A).toarray(),
+                               rtol=tol, atol=tol)
+
+
+def test_gpu_spgemm_mid_bin():
+    # rows with expansion bound in (128, 1024]: ~30 nnz/row squared
+    A = banded_matrix(400, nnz_per_row=31)
+    C = lsp.csr_array(A) @ lsp.csr_array(A)
+    np.testing.assert_allclose(to_np(C.todense()), (A @ A).toarray(),
+                               rtol=1e-12, atol=1e-12)
+
+
+def test_gpu_spgemm_large_bin():
+    # expansion bound in (1024, 4096]
+    A = banded_matrix(600, nnz_per_row=57)
+    C = lsp.csr_array(A) @ lsp.csr_array(A)
+    np.testing.assert_allclose(to_np(C.todense()), (A @ A).toarray(),
+                               rtol=1e-12, atol=1e-12)
+
+
+def test_gpu_spgemm_global_bin():
+    # a fully dense row drives ub ~ n*avg > 4096 -> HBM hash fallback
+    n = 700
+    D = (np.random.default_rng(6).random((n, n)) > 0.98).astype(float)
+    D[0, :] = 1.0
+    D[:, 0] = 1.0
+    S = sp.csr_array(D)
+    C = lsp.csr_array(S) @ lsp.csr_array(S)
+    np.testing.assert_allclose(to_np(C.todense()), (S @ S).toarray(),
+                               rtol=1e-12, atol=1e-12)
+
+
+def test_gpu_spgemm_mixed_bins():
+    # matrix whose rows span all four bins in one call
+    n = 800
+    rng = np.random.default_rng(7)
+    rows, cols, vals = [], [], []
+    for i in range(n):
+        if i < 4:
+            nnz = n  # dense row -> global bin
+        elif i < 40:
+            nnz = 60
+        elif i < 400:
+            nnz = 12
+        else:
+            nnz = 3
+        c = rng.choice(n, size=min(nnz, n), replace=False)
+        rows.extend([i] * len(c))
+        cols.extend(c.tolist())
+        vals.extend(rng.standard_normal(len(c)).tolist())
+    S = sp.csr_array((vals, (rows, cols)), shape=(n, n))
+    C = lsp.csr_array(S) @ lsp.csr_array(S)
+    np.testing.assert_allclose(to_np(C.todense()), (S @ S).toarray(),
+                               rtol=1e-11, atol=1e-11)
+
+
+@pytest.mark.parametrize("isalpha", [True, False])
+@pytest.mark.parametrize("negate", [True, False])
+@pytest.mark.parametrize("dtype", [np.float64, np.complex128])
+def test_gpu_axpby(isalpha, negate, dtype):
+    from legate_sparse.linalg import cg_axpby
+    rng = np.random.default_rng(8)
+    n = 10000
+
+    def mk(seed):
+        v = rng.standard_normal(n)
+        if np.dtype(dtype).kind == "c":
+            v = v + 1j * rng.standard_normal(n)
+        return v.astype(dtype)
+
+    y0, x0 = mk(1), mk(2)
+    a = np.array([1.7], dtype=dtype)
+    b = np.array([-0.4], dtype=dtype)
+    y = torch.from_numpy(y0.copy()).cuda()
+    x = torch.from_numpy(x0.copy()).cuda()
+    cg_axpby(y, x, torch.from_numpy(a).cuda(), torch.from_numpy(b).cuda(),
+             isalpha=isalpha, negate=negate)
+    val = (-1 if negate else 1) * a[0] / b[0]
+    want = val * x0 + y0 if isalpha else x0 + val * y0
+    np.testing.assert_allclose(to_np(y), want, rtol=1e-12)
+
+
+@pytest.mark.parametrize("dtype", [np.float32, np.float64, np.complex128])
+def test_gpu_vdot(dtype):
+    from legate_sparse import ops
+    rng = np.random.default_rng(9)
+    n = 100003
+    x = rng.standard_normal(n)
+    y = rng.standard_normal(n)
+    if np.dtype(dtype).kind == "c":
+        x = x + 1j * rng.standard_normal(n)
+        y = y + 1j * rng.standard_normal(n)
+    x = x.astype(dtype)
+    y = y.astype(dtype)
+    got = ops.vdot(torch.from_numpy(x).cuda(), torch.from_numpy(y).cuda())
+    want = np.vdot(x, y)
+    tol = 1e-3 if np.dtype(dtype).itemsize <= 8 else 1e-10
+    np.testing.assert_allclose(to_np(got)[0], want, rtol=tol)
+
+
+def test_gpu_dense_csr_roundtrip():
+    rng = np.random.default_rng(10)
+    D = rng.standard_normal((123, 77))
+    D[rng.random((123, 77)) < 0.6] = 0.0
+    A = lsp.csr_array(torch.from_numpy(D).cuda())
+    S = sp.csr_array(D)
+    assert A.nnz == S.nnz
+    np.testing.assert_array_equal(to_np(A.indptr), S.indptr)
+    np.testing.assert_array_equal(to_np(A.indices), S.indices)
+    np.testing.assert_allclose(to_np(A.todense()), D)
+
+
+def test_gpu_diagonal():
+    S = sample_csr(211, 211, 0.05, seed=11)
+    A = lsp.csr_array(S)
+    np.testing.assert_allclose(to_np(A.diagonal()), S.diagonal())
+
+
+def test_gpu_transpose():
+    S = sample_csr(150, 90, 0.1, seed=12)
+    A = lsp.csr_array(S)
+    np.testing.assert_allclose(to_np(A.T.todense()), S.T.toarray())
+
+
+def test_gpu_cg_poisson():
+    from legate_sparse.gallery import poisson_2d
+    A = poisson_2d(64, 64)
+    assert A.data.is_cuda
+    b = torch.ones(A.shape[0], dtype=torch.float64, device="cuda")
+    x, info = lsp.linalg.cg(A, b, rtol=1e-10, maxiter=2000)
+    assert info == 0
+    r = to_np(b - A @ x)
+    assert np.linalg.norm(r) <= 1e-8 * np.linalg.norm(to_np(b))
+
+
+def test_gpu_gmres_poisson():
+    from legate_sparse.gallery import poisson_2d
+    A = poisson_2d(32, 32)
+    b = torch.ones(A.shape[0], dtype=torch.float64, device="cuda")
+    x, info = lsp.linalg.gmres(A, b, rtol=1e-8, restart=50, maxiter=3000)
+    assert info == 0
+
+
+def test_gpu_ufuncs_and_scalars():
+    S = sample_csr(100, 100, 0.05, seed=13)
+    A = lsp.csr_array(S)
+    got = (A * 2.0).tanh()
+    want = S.copy()
+    want.data = np.tanh(2.0 * S.data)
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray(),
+                               rtol=1e-12)

@@ -1,0 +1,67 @@
+# SPDX-License-Identifier: Apache-2.0
+"""mmread vs scipy.io.mmread (reference test_io.py semantics)."""
+import io
+
+import numpy as np
+import pytest
+import scipy.io
+import scipy.sparse as sp
+
+import legate_sparse as lsp
+from testutils import sample_csr, to_np
+
+
+def _roundtrip(S, tmp_path, name, symmetry=None):
+    path = str(tmp_path / name)
+    scipy.io.mmwrite(path, S, symmetry=symmetry)
+    got = lsp.mmread(path)
+    want = scipy.io.mmread(path).tocsr()
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray())
+    assert got.shape == want.shape
+
+
+def test_mmread_general(tmp_path):
+    _roundtrip(sample_csr(12, 9, 0.3, seed=1), tmp_path, "gen.mtx")
+
+
+def test_mmread_symmetric(tmp_path):
+    S = sample_csr(10, 10, 0.3, seed=2)
+    S = 0.5 * (S + S.T)
+    _roundtrip(S.tocsr(), tmp_path, "sym.mtx", symmetry="symmetric")
+
+
+def test_mmread_pattern(tmp_path):
+    S = sample_csr(8, 8, 0.4, seed=3)
+    S.data[:] = 1.0
+    path = str(tmp_path / "pat.mtx")
+    # write a pattern file by hand
+    coo = S.tocoo()
+    with open(path, "w") as f:
+        f.write("%%MatrixMarket matrix coordinate pattern general\n")
+        f.write(f"{S.shape[0]} {S.shape[1]} {S.nnz}\n")
+        for r, c in zip(coo.row, coo.col):
+            f.write(f"{r + 1} {c + 1}\n")
+    got = lsp.mmread(path)
+    want = scipy.io.mmread(path).tocsr()
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray())
+
+
+def test_mmread_complex(tmp_path):
+    S = sample_csr(7, 7, 0.4, seed=4, dtype=np.complex128)
+    _roundtrip(S, tmp_path, "cplx.mtx")
+
+
+def test_mmwrite_roundtrip(tmp_path):
+    S = sample_csr(9, 9, 0.3, seed=5)
+    path = str(tmp_path / "rt.mtx")
+    lsp.mmwrite(path, lsp.csr_array(S))
+    want = scipy.io.mmread(path).tocsr()
+    np.testing.assert_allclose(want.toarray(), S.toarray())
+
+
+def test_mmread_bad_banner(tmp_path):
+    path = str(tmp_path / "bad.mtx")
+    with open(path, "w") as f:
+        f.write("not a matrix market file\n1 1 1\n1 1 1.0\n")
+    with pytest.raises(ValueError):
+        lsp.mmread(path)
