@@ -40,6 +40,13 @@ from .types import (
 _INDEX_DTYPE = torch.int64
 
 
+def _index_dtype_for(shape) -> torch.dtype:
+    """int32 column indices whenever the matrix allows — halves the
+    streamed index bytes of the memory-bound SpMV/SpGEMM kernels."""
+    return torch.int32 if max(int(shape[0]), int(shape[1])) < 2**31 - 1 \
+        else torch.int64
+
+
 def _default_device() -> torch.device:
     return runtime.device
 
@@ -150,7 +157,8 @@ class csr_array(CompressedBase):
         part = runtime.partition(M)
         lo, hi = part.lo(runtime.rank), part.hi(runtime.rank)
         local = dense[lo:hi].to(device)
-        indptr, indices, vals = ops.dense_to_csr(local)
+        indptr, indices, vals = ops.dense_to_csr(
+            local, idx_dtype=_index_dtype_for((M, N)))
         self._init_local(indptr, indices, vals, (M, N))
 
     def _init_local(self, indptr, indices, data, shape):
@@ -164,7 +172,8 @@ class csr_array(CompressedBase):
         assert indptr.numel() == (self._row_hi - self._row_lo) + 1, (
             indptr.numel(), self._row_lo, self._row_hi)
         self._indptr = indptr.contiguous()
-        self._indices = indices.contiguous()
+        idt = _index_dtype_for(self._shape)
+        self._indices = indices.to(idt).contiguous()
         self._data = data.contiguous()
         self._global_nnz: Optional[int] = None
 
@@ -391,7 +400,7 @@ class csr_array(CompressedBase):
         if axis == 0:
             out = torch.zeros(self._shape[1], dtype=self._data.dtype,
                               device=self._data.device)
-            out.scatter_add_(0, self._indices, self._data)
+            out.scatter_add_(0, self._indices.long(), self._data)
             if runtime.world_size > 1:
                 comm.allreduce_(out)
             return out
@@ -409,7 +418,7 @@ class csr_array(CompressedBase):
         grow = torch.repeat_interleave(
             torch.arange(self._row_lo, self._row_hi, device=dev),
             self._indptr[1:] - self._indptr[:-1])
-        col = self._indices
+        col = self._indices.long()
         val = self._data
         newpart = runtime.partition(N)
         if runtime.world_size > 1:
@@ -419,7 +428,7 @@ class csr_array(CompressedBase):
             order = torch.argsort(owner, stable=True)
             grow, col, val, owner = (grow[order], col[order], val[order],
                                      owner[order])
-            counts = torch.bincount(owner.cpu(),
+            counts = torch.bincount(owner.cpu().long(),
                                     minlength=runtime.world_size).tolist()
             splits_r = list(torch.split(grow, counts))
             splits_c = list(torch.split(col, counts))
@@ -515,8 +524,12 @@ def spgemm_csr_csr_csr(A: csr_array, B: csr_array) -> csr_array:
     if B._data.dtype != dt:
         B = B.astype(to_numpy_dtype(dt))
     B_indptr, B_indices, B_data = _gather_full_B(B)
+    A_indices = A._indices
+    if A_indices.dtype != B_indices.dtype:  # unify for the kernels
+        A_indices = A_indices.to(torch.int64)
+        B_indices = B_indices.to(torch.int64)
     C_indptr, C_indices, C_vals = ops.spgemm_local(
-        A._indptr, A._indices, A._data, B_indptr, B_indices, B_data,
+        A._indptr, A_indices, A._data, B_indptr, B_indices, B_data,
         B.shape[1])
     out = csr_array.__new__(csr_array)
     out._init_local(C_indptr, C_indices, C_vals, (A.shape[0], B.shape[1]))

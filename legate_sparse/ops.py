@@ -26,9 +26,15 @@ _DTYPE_CODE = {
     torch.complex128: 3,
 }
 
+_IDX_CODE = {torch.int32: 0, torch.int64: 1}
+
 
 def _code(t: torch.Tensor) -> int:
     return _DTYPE_CODE[t.dtype]
+
+
+def _icode(t: torch.Tensor) -> int:
+    return _IDX_CODE[t.dtype]
 
 
 def _use_hip(t: torch.Tensor) -> bool:
@@ -53,15 +59,16 @@ def spmv(indptr: torch.Tensor, indices: torch.Tensor, vals: torch.Tensor,
         ext = _cext.require_hip()
         ext.spmv(indptr.data_ptr(), indices.data_ptr(), vals.data_ptr(),
                  x.data_ptr(), y.data_ptr(), n_rows, vals.numel(),
-                 _code(vals), accumulate, _stream())
+                 _code(vals), _icode(indices), accumulate, _stream())
         return y
     if not vals.is_cuda and _cext.has_cpu():
         _cext.require_cpu().spmv(indptr.data_ptr(), indices.data_ptr(),
                                  vals.data_ptr(), x.data_ptr(), y.data_ptr(),
-                                 n_rows, _code(vals), accumulate)
+                                 n_rows, _code(vals), _icode(indices),
+                                 accumulate)
         return y
     # torch fallback (debug / extension-less CPU)
-    prod = vals * x[indices]
+    prod = vals * x[indices.long()]
     row_ids = torch.repeat_interleave(
         torch.arange(n_rows, device=vals.device),
         (indptr[1:] - indptr[:-1]),
@@ -92,18 +99,18 @@ def spgemm_local(
         ext.spgemm_symbolic(A_indptr.data_ptr(), A_indices.data_ptr(),
                             n_rows, B_indptr.data_ptr(),
                             B_indices.data_ptr(), int(n_colsB),
-                            row_nnz.data_ptr())
+                            row_nnz.data_ptr(), _icode(A_indices))
         C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64)
         torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
         nnz = int(C_indptr[-1])
-        C_indices = torch.empty(nnz, dtype=torch.int64)
+        C_indices = torch.empty(nnz, dtype=A_indices.dtype)
         C_vals = torch.empty(nnz, dtype=A_vals.dtype)
         ext.spgemm_numeric(A_indptr.data_ptr(), A_indices.data_ptr(),
                            A_vals.data_ptr(), n_rows, B_indptr.data_ptr(),
                            B_indices.data_ptr(), B_vals.data_ptr(),
                            int(n_colsB), C_indptr.data_ptr(),
                            C_indices.data_ptr(), C_vals.data_ptr(),
-                           _code(A_vals))
+                           _code(A_vals), _icode(A_indices))
         return C_indptr, C_indices, C_vals
     # torch fallback: ESC (expand - sort - compress), works on any device.
     return _spgemm_esc(A_indptr, A_indices, A_vals, B_indptr, B_indices,
@@ -115,26 +122,30 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     """Binned hash-table Gustavson on gfx950 (src/hip/spgemm.hip).
 
     Rows are binned by their expansion upper bound so the per-row hash
-    table fits LDS; oversize rows fall back to an HBM workspace table
-    (the automatic analogue of the reference's cuSPARSE ALG1/ALG3 switch,
-    spgemm_csr_csr_csr.cu:196-216)."""
+    table fits LDS (sorted output via in-LDS bitonic); oversize rows fall
+    back to an HBM workspace table (the automatic analogue of the
+    reference's cuSPARSE ALG1/ALG3 switch, spgemm_csr_csr_csr.cu:196-216)
+    and are canonically sorted afterwards."""
     ext = _cext.require_hip()
     dev = A_vals.device
     st = _stream()
     n_rows = A_indptr.numel() - 1
     code = _code(A_vals)
+    icode = _icode(A_indices)
+    assert A_indices.dtype == B_indices.dtype
     if n_colsB >= 2 ** 32:
         raise NotImplementedError("SpGEMM requires B.shape[1] < 2^32")
 
     ub = torch.empty(n_rows, dtype=torch.int64, device=dev)
     ext.spgemm_row_ub(A_indptr.data_ptr(), A_indices.data_ptr(),
-                      B_indptr.data_ptr(), ub.data_ptr(), n_rows, st)
-    bins = [
-        torch.nonzero(ub <= 128).reshape(-1),
-        torch.nonzero((ub > 128) & (ub <= 1024)).reshape(-1),
-        torch.nonzero((ub > 1024) & (ub <= 4096)).reshape(-1),
-    ]
-    rows_g = torch.nonzero(ub > 4096).reshape(-1)
+                      B_indptr.data_ptr(), ub.data_ptr(), n_rows, icode, st)
+    cuts = ext.spgemm_lds_bins  # (48, 128, 1024, 4096)
+    bins = []
+    prev = -1
+    for c in cuts:
+        bins.append(torch.nonzero((ub > prev) & (ub <= c)).reshape(-1))
+        prev = c
+    rows_g = torch.nonzero(ub > cuts[-1]).reshape(-1)
 
     row_nnz = torch.zeros(n_rows, dtype=torch.int64, device=dev)
     for cfg, rows in enumerate(bins):
@@ -142,7 +153,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
             ext.spgemm_symbolic_lds(
                 cfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
                 A_indices.data_ptr(), B_indptr.data_ptr(),
-                B_indices.data_ptr(), row_nnz.data_ptr(), st)
+                B_indices.data_ptr(), row_nnz.data_ptr(), icode, st)
     g_keys = g_off = g_sizes = None
     if rows_g.numel():
         ub_g = ub[rows_g]
@@ -157,12 +168,12 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
             rows_g.data_ptr(), rows_g.numel(), A_indptr.data_ptr(),
             A_indices.data_ptr(), B_indptr.data_ptr(),
             B_indices.data_ptr(), g_keys.data_ptr(), g_off.data_ptr(),
-            g_sizes.data_ptr(), row_nnz.data_ptr(), st)
+            g_sizes.data_ptr(), row_nnz.data_ptr(), icode, st)
 
     C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
     torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
     nnz = int(C_indptr[-1].item())
-    C_indices = torch.empty(nnz, dtype=torch.int64, device=dev)
+    C_indices = torch.empty(nnz, dtype=A_indices.dtype, device=dev)
     C_vals = torch.empty(nnz, dtype=A_vals.dtype, device=dev)
 
     for cfg, rows in enumerate(bins):
@@ -172,7 +183,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 A_indices.data_ptr(), A_vals.data_ptr(),
                 B_indptr.data_ptr(), B_indices.data_ptr(),
                 B_vals.data_ptr(), C_indptr.data_ptr(),
-                C_indices.data_ptr(), C_vals.data_ptr(), code, st)
+                C_indices.data_ptr(), C_vals.data_ptr(), code, icode, st)
     if rows_g.numel():
         g_keys.fill_(-1)
         g_vals = torch.zeros(g_keys.numel(), dtype=A_vals.dtype, device=dev)
@@ -182,14 +193,23 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
             B_indices.data_ptr(), B_vals.data_ptr(), g_keys.data_ptr(),
             g_vals.data_ptr(), g_off.data_ptr(), g_sizes.data_ptr(),
             C_indptr.data_ptr(), C_indices.data_ptr(), C_vals.data_ptr(),
-            code, st)
-
-    # canonical per-row column sort (hash compaction emits unsorted)
-    row_ids = torch.repeat_interleave(
-        torch.arange(n_rows, device=dev), C_indptr[1:] - C_indptr[:-1])
-    key = row_ids * int(n_colsB) + C_indices
-    order = torch.argsort(key)
-    return C_indptr, C_indices[order], C_vals[order]
+            code, icode, st)
+        # canonical per-row sort for the (rare) global-bin rows only
+        cnts = (C_indptr[rows_g + 1] - C_indptr[rows_g])
+        total_e = int(cnts.sum().item())
+        if total_e:
+            starts = C_indptr[rows_g]
+            seg = torch.repeat_interleave(
+                torch.arange(rows_g.numel(), device=dev), cnts)
+            seg_start = torch.cumsum(cnts, 0) - cnts
+            pos = torch.arange(total_e, device=dev) - torch.repeat_interleave(
+                seg_start, cnts)
+            ent = torch.repeat_interleave(starts, cnts) + pos
+            keys = seg * int(n_colsB) + C_indices[ent].long()
+            order = torch.argsort(keys)
+            C_indices[ent] = C_indices[ent][order]
+            C_vals[ent] = C_vals[ent][order]
+    return C_indptr, C_indices, C_vals
 
 
 def _spgemm_esc(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
@@ -199,7 +219,7 @@ def _spgemm_esc(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     a_rows = torch.repeat_interleave(
         torch.arange(n_rows, device=dev), A_indptr[1:] - A_indptr[:-1])
     k = A_indices
-    blen = (B_indptr[1:] - B_indptr[:-1])[k]
+    blen = (B_indptr[1:] - B_indptr[:-1])[k.long()]
     e_rows = torch.repeat_interleave(a_rows, blen)
     e_avals = torch.repeat_interleave(A_vals, blen)
     total = int(blen.sum())
@@ -207,10 +227,10 @@ def _spgemm_esc(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     seg_starts = torch.cumsum(blen, 0) - blen
     pos = torch.arange(total, device=dev) - torch.repeat_interleave(
         seg_starts, blen)
-    b_off = torch.repeat_interleave(B_indptr[k], blen) + pos
+    b_off = torch.repeat_interleave(B_indptr[k.long()], blen) + pos
     e_cols = B_indices[b_off]
     e_vals = e_avals * B_vals[b_off]
-    key = e_rows * int(n_colsB) + e_cols
+    key = e_rows * int(n_colsB) + e_cols.long()
     key_sorted, order = torch.sort(key, stable=True)
     vals_sorted = e_vals[order]
     uniq, inverse = torch.unique_consecutive(key_sorted, return_inverse=True)
@@ -233,15 +253,15 @@ def csr_to_dense(indptr, indices, vals, n_rows, n_cols) -> torch.Tensor:
         ext = _cext.require_hip()
         ext.csr_to_dense(indptr.data_ptr(), indices.data_ptr(),
                          vals.data_ptr(), out.data_ptr(), n_rows, n_cols,
-                         _code(vals), _stream())
+                         _code(vals), _icode(indices), _stream())
         return out
     row_ids = torch.repeat_interleave(
         torch.arange(n_rows, device=vals.device), indptr[1:] - indptr[:-1])
-    out[row_ids, indices] = vals
+    out[row_ids, indices.long()] = vals
     return out
 
 
-def dense_to_csr(dense: torch.Tensor):
+def dense_to_csr(dense: torch.Tensor, idx_dtype=torch.int64):
     n_rows, n_cols = dense.shape
     dense = dense.contiguous()
     if _use_hip(dense):
@@ -253,18 +273,19 @@ def dense_to_csr(dense: torch.Tensor):
                              device=dense.device)
         torch.cumsum(row_nnz, dim=0, out=indptr[1:])
         nnz = int(indptr[-1].item())
-        indices = torch.empty(nnz, dtype=torch.int64, device=dense.device)
+        indices = torch.empty(nnz, dtype=idx_dtype, device=dense.device)
         vals = torch.empty(nnz, dtype=dense.dtype, device=dense.device)
         ext.dense_to_csr_fill(dense.data_ptr(), indptr.data_ptr(),
                               indices.data_ptr(), vals.data_ptr(), n_rows,
-                              n_cols, _code(dense), _stream())
+                              n_cols, _code(dense), _IDX_CODE[idx_dtype],
+                              _stream())
         return indptr, indices, vals
     mask = dense != 0
     row_nnz = mask.sum(dim=1)
     indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dense.device)
     torch.cumsum(row_nnz, dim=0, out=indptr[1:])
     nz = mask.nonzero(as_tuple=True)
-    return indptr, nz[1].to(torch.int64), dense[nz]
+    return indptr, nz[1].to(idx_dtype), dense[nz]
 
 
 def diagonal(indptr, indices, vals, n_rows, row_offset: int) -> torch.Tensor:
@@ -275,11 +296,11 @@ def diagonal(indptr, indices, vals, n_rows, row_offset: int) -> torch.Tensor:
         ext = _cext.require_hip()
         ext.diagonal(indptr.data_ptr(), indices.data_ptr(), vals.data_ptr(),
                      out.data_ptr(), n_rows, int(row_offset), _code(vals),
-                     _stream())
+                     _icode(indices), _stream())
         return out
     row_ids = torch.repeat_interleave(
         torch.arange(n_rows, device=vals.device), indptr[1:] - indptr[:-1])
-    hit = indices == (row_ids + row_offset)
+    hit = indices.long() == (row_ids + row_offset)
     out[row_ids[hit]] = vals[hit]
     return out
 

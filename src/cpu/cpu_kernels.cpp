@@ -6,11 +6,12 @@
 //                               spmv_omp.cc:36-45)
 //   - Gustavson SpGEMM 2-phase (reference spgemm_csr_csr_csr.cc:38-160,
 //                               spgemm_csr_csr_csr_omp.cc:53-167)
-// re-designed: plain indptr instead of Legion pos rects, int64 indices,
-// dense per-thread accumulators, OpenMP dynamic row scheduling.
+// re-designed: plain indptr instead of Legion pos rects, int32/int64
+// indices, dense per-thread accumulators, OpenMP dynamic row scheduling.
 //
 // All pointers arrive as uintptr_t from torch tensors (contiguous,
-// host-resident). dtype codes: 0=f32, 1=f64, 2=c64, 3=c128.
+// host-resident). value dtype codes: 0=f32, 1=f64, 2=c64, 3=c128;
+// index dtype codes: 0=int32, 1=int64.
 
 #include <pybind11/pybind11.h>
 #include <algorithm>
@@ -27,8 +28,8 @@ namespace py = pybind11;
 
 using i64 = int64_t;
 
-template <typename T>
-static void spmv_impl(const i64* indptr, const i64* indices, const T* vals,
+template <typename T, typename I>
+static void spmv_impl(const i64* indptr, const I* indices, const T* vals,
                       const T* x, T* y, i64 n_rows, bool accumulate) {
 #pragma omp parallel for schedule(dynamic, 256)
   for (i64 i = 0; i < n_rows; ++i) {
@@ -41,9 +42,10 @@ static void spmv_impl(const i64* indptr, const i64* indices, const T* vals,
 }
 
 // Gustavson symbolic: row_nnz[i] = |union of B-row col sets over A's row i|.
-static void spgemm_symbolic_impl(const i64* A_indptr, const i64* A_indices,
+template <typename I>
+static void spgemm_symbolic_impl(const i64* A_indptr, const I* A_indices,
                                  i64 n_rowsA, const i64* B_indptr,
-                                 const i64* B_indices, i64 n_colsB,
+                                 const I* B_indices, i64 n_colsB,
                                  i64* row_nnz) {
 #pragma omp parallel
   {
@@ -52,9 +54,9 @@ static void spgemm_symbolic_impl(const i64* A_indptr, const i64* A_indices,
     for (i64 i = 0; i < n_rowsA; ++i) {
       i64 count = 0;
       for (i64 jp = A_indptr[i]; jp < A_indptr[i + 1]; ++jp) {
-        const i64 k = A_indices[jp];
+        const i64 k = (i64)A_indices[jp];
         for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-          const i64 col = B_indices[kp];
+          const i64 col = (i64)B_indices[kp];
           if (marker[col] != i) {
             marker[col] = i;
             ++count;
@@ -69,12 +71,12 @@ static void spgemm_symbolic_impl(const i64* A_indptr, const i64* A_indices,
 // Gustavson numeric: emits columns in sorted order (unlike the reference,
 // which emits insertion order — spgemm_csr_csr_csr.cc:92-160; sorted output
 // keeps scipy compatibility bit-clean).
-template <typename T>
-static void spgemm_numeric_impl(const i64* A_indptr, const i64* A_indices,
+template <typename T, typename I>
+static void spgemm_numeric_impl(const i64* A_indptr, const I* A_indices,
                                 const T* A_vals, i64 n_rowsA,
-                                const i64* B_indptr, const i64* B_indices,
+                                const i64* B_indptr, const I* B_indices,
                                 const T* B_vals, i64 n_colsB,
-                                const i64* C_indptr, i64* C_indices,
+                                const i64* C_indptr, I* C_indices,
                                 T* C_vals) {
 #pragma omp parallel
   {
@@ -85,10 +87,10 @@ static void spgemm_numeric_impl(const i64* A_indptr, const i64* A_indices,
     for (i64 i = 0; i < n_rowsA; ++i) {
       cols.clear();
       for (i64 jp = A_indptr[i]; jp < A_indptr[i + 1]; ++jp) {
-        const i64 k = A_indices[jp];
+        const i64 k = (i64)A_indices[jp];
         const T a = A_vals[jp];
         for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-          const i64 col = B_indices[kp];
+          const i64 col = (i64)B_indices[kp];
           if (marker[col] != i) {
             marker[col] = i;
             workspace[col] = a * B_vals[kp];
@@ -101,7 +103,7 @@ static void spgemm_numeric_impl(const i64* A_indptr, const i64* A_indices,
       std::sort(cols.begin(), cols.end());
       i64 out = C_indptr[i];
       for (i64 col : cols) {
-        C_indices[out] = col;
+        C_indices[out] = (I)col;
         C_vals[out] = workspace[col];
         ++out;
       }
@@ -120,50 +122,59 @@ enum DtypeCode { F32 = 0, F64 = 1, C64 = 2, C128 = 3 };
     default: throw std::runtime_error("bad dtype code");             \
   }
 
+#define DISPATCH_IDX(code, CALL)                         \
+  switch (code) {                                        \
+    case 0: { using index_t = int32_t; CALL; break; }    \
+    case 1: { using index_t = int64_t; CALL; break; }    \
+    default: throw std::runtime_error("bad index dtype code"); \
+  }
+
 static void spmv(uintptr_t indptr, uintptr_t indices, uintptr_t vals,
                  uintptr_t x, uintptr_t y, i64 n_rows, int dtype,
-                 bool accumulate) {
-  DISPATCH_VAL(dtype, spmv_impl<scalar_t>(
+                 int idx_dtype, bool accumulate) {
+  DISPATCH_VAL(dtype, DISPATCH_IDX(idx_dtype, (spmv_impl<scalar_t, index_t>(
       reinterpret_cast<const i64*>(indptr),
-      reinterpret_cast<const i64*>(indices),
+      reinterpret_cast<const index_t*>(indices),
       reinterpret_cast<const scalar_t*>(vals),
       reinterpret_cast<const scalar_t*>(x),
-      reinterpret_cast<scalar_t*>(y), n_rows, accumulate));
+      reinterpret_cast<scalar_t*>(y), n_rows, accumulate))));
 }
 
 static void spgemm_symbolic(uintptr_t A_indptr, uintptr_t A_indices,
                             i64 n_rowsA, uintptr_t B_indptr,
                             uintptr_t B_indices, i64 n_colsB,
-                            uintptr_t row_nnz) {
-  spgemm_symbolic_impl(reinterpret_cast<const i64*>(A_indptr),
-                       reinterpret_cast<const i64*>(A_indices), n_rowsA,
-                       reinterpret_cast<const i64*>(B_indptr),
-                       reinterpret_cast<const i64*>(B_indices), n_colsB,
-                       reinterpret_cast<i64*>(row_nnz));
+                            uintptr_t row_nnz, int idx_dtype) {
+  DISPATCH_IDX(idx_dtype, (spgemm_symbolic_impl<index_t>(
+      reinterpret_cast<const i64*>(A_indptr),
+      reinterpret_cast<const index_t*>(A_indices), n_rowsA,
+      reinterpret_cast<const i64*>(B_indptr),
+      reinterpret_cast<const index_t*>(B_indices), n_colsB,
+      reinterpret_cast<i64*>(row_nnz))));
 }
 
 static void spgemm_numeric(uintptr_t A_indptr, uintptr_t A_indices,
                            uintptr_t A_vals, i64 n_rowsA, uintptr_t B_indptr,
                            uintptr_t B_indices, uintptr_t B_vals, i64 n_colsB,
                            uintptr_t C_indptr, uintptr_t C_indices,
-                           uintptr_t C_vals, int dtype) {
-  DISPATCH_VAL(dtype, spgemm_numeric_impl<scalar_t>(
-      reinterpret_cast<const i64*>(A_indptr),
-      reinterpret_cast<const i64*>(A_indices),
-      reinterpret_cast<const scalar_t*>(A_vals), n_rowsA,
-      reinterpret_cast<const i64*>(B_indptr),
-      reinterpret_cast<const i64*>(B_indices),
-      reinterpret_cast<const scalar_t*>(B_vals), n_colsB,
-      reinterpret_cast<const i64*>(C_indptr),
-      reinterpret_cast<i64*>(C_indices),
-      reinterpret_cast<scalar_t*>(C_vals)));
+                           uintptr_t C_vals, int dtype, int idx_dtype) {
+  DISPATCH_VAL(dtype, DISPATCH_IDX(idx_dtype, (
+      spgemm_numeric_impl<scalar_t, index_t>(
+          reinterpret_cast<const i64*>(A_indptr),
+          reinterpret_cast<const index_t*>(A_indices),
+          reinterpret_cast<const scalar_t*>(A_vals), n_rowsA,
+          reinterpret_cast<const i64*>(B_indptr),
+          reinterpret_cast<const index_t*>(B_indices),
+          reinterpret_cast<const scalar_t*>(B_vals), n_colsB,
+          reinterpret_cast<const i64*>(C_indptr),
+          reinterpret_cast<index_t*>(C_indices),
+          reinterpret_cast<scalar_t*>(C_vals)))));
 }
 
 PYBIND11_MODULE(_cpu_kernels, m) {
   m.doc() = "legate_sparse CPU/OpenMP kernels";
   m.def("spmv", &spmv, py::arg("indptr"), py::arg("indices"), py::arg("vals"),
         py::arg("x"), py::arg("y"), py::arg("n_rows"), py::arg("dtype"),
-        py::arg("accumulate") = false);
+        py::arg("idx_dtype"), py::arg("accumulate") = false);
   m.def("spgemm_symbolic", &spgemm_symbolic);
   m.def("spgemm_numeric", &spgemm_numeric);
 #ifdef _OPENMP

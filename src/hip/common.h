@@ -88,6 +88,16 @@ __device__ inline Cplx<double> ls_zero<Cplx<double>>() { return {0., 0.}; }
     default: throw std::runtime_error("bad dtype code");            \
   }
 
+// index dtype codes shared with Python: 0=int32, 1=int64
+// (int32 column indices halve the streamed index bytes of SpMV/SpGEMM —
+// the dominant optimization for memory-bound fp64 sparse kernels)
+#define DISPATCH_IDX_T(code, CALL)                                  \
+  switch (code) {                                                   \
+    case 0: { using idx_t = int32_t; CALL; break; }                 \
+    case 1: { using idx_t = int64_t; CALL; break; }                 \
+    default: throw std::runtime_error("bad index dtype code");      \
+  }
+
 // ---------------------------------------------------------------------------
 // Wave / block reductions (64-wide wave; __shfl_down works across 64 lanes)
 // ---------------------------------------------------------------------------

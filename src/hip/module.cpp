@@ -12,29 +12,31 @@ namespace py = pybind11;
 using i64 = int64_t;
 
 void ls_spmv(uintptr_t, uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64,
-             i64, int, bool, uintptr_t);
-void ls_spgemm_row_ub(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64,
+             i64, int, int, bool, uintptr_t);
+void ls_spgemm_row_ub(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, int,
                       uintptr_t);
 void ls_spgemm_symbolic_lds(int, uintptr_t, i64, uintptr_t, uintptr_t,
-                            uintptr_t, uintptr_t, uintptr_t, uintptr_t);
+                            uintptr_t, uintptr_t, uintptr_t, int,
+                            uintptr_t);
 void ls_spgemm_numeric_lds(int, uintptr_t, i64, uintptr_t, uintptr_t,
                            uintptr_t, uintptr_t, uintptr_t, uintptr_t,
-                           uintptr_t, uintptr_t, uintptr_t, int, uintptr_t);
+                           uintptr_t, uintptr_t, uintptr_t, int, int,
+                           uintptr_t);
 void ls_spgemm_symbolic_global(uintptr_t, i64, uintptr_t, uintptr_t,
                                uintptr_t, uintptr_t, uintptr_t, uintptr_t,
-                               uintptr_t, uintptr_t, uintptr_t);
+                               uintptr_t, uintptr_t, int, uintptr_t);
 void ls_spgemm_numeric_global(uintptr_t, i64, uintptr_t, uintptr_t,
                               uintptr_t, uintptr_t, uintptr_t, uintptr_t,
                               uintptr_t, uintptr_t, uintptr_t, uintptr_t,
-                              uintptr_t, uintptr_t, uintptr_t, int,
+                              uintptr_t, uintptr_t, uintptr_t, int, int,
                               uintptr_t);
 void ls_csr_to_dense(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, i64,
-                     int, uintptr_t);
+                     int, int, uintptr_t);
 void ls_dense_to_csr_nnz(uintptr_t, uintptr_t, i64, i64, int, uintptr_t);
 void ls_dense_to_csr_fill(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64,
-                          i64, int, uintptr_t);
+                          i64, int, int, uintptr_t);
 void ls_diagonal(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, i64, int,
-                 uintptr_t);
+                 int, uintptr_t);
 void ls_axpby(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, bool, bool,
               int, uintptr_t);
 void ls_vdot(uintptr_t, uintptr_t, uintptr_t, i64, bool, int, uintptr_t);
@@ -54,4 +56,5 @@ PYBIND11_MODULE(_hip_kernels, m) {
   m.def("axpby", &ls_axpby);
   m.def("vdot", &ls_vdot);
   m.attr("arch") = "gfx950";
+  m.attr("spgemm_lds_bins") = py::make_tuple(48, 128, 1024, 4096);
 }

@@ -5,22 +5,23 @@
 // (spgemm_csr_csr_csr.cu:196-332) with a 2-phase LDS-hash design:
 //
 //   phase 0  row_upper_bound: ub[i] = sum of B-row lengths over A's row i
-//            (cheap; drives the size binning below)
-//   phase 1  symbolic: per C-row distinct-column count via hash insert
-//   phase 2  numeric:  hash insert + accumulate, then compact
+//   phase 1  symbolic: per C-row distinct-column count via LDS hash insert
+//   phase 2  numeric:  hash insert + accumulate, in-LDS bitonic sort,
+//            coalesced sorted write-out (no global sort pass — profiling
+//            showed a rocprim canonical sort costing more than the numeric
+//            phase itself)
 //
 // Rows are binned by ub so the hash table fits LDS (160 KiB/CU):
-//   cfg0: ub <=  128 — 16 lanes/row, 16 rows/block, 256-slot table
-//   cfg1: ub <= 1024 — 1 wave/row,    2 rows/block, 2048-slot table
-//   cfg2: ub <= 4096 — 1 block/row,                 8192-slot table
-//   global bin (ub > 4096): per-row power-of-two table in HBM workspace
-//     (the ALG3-style memory fallback; reference picks cuSPARSE ALG1/ALG3
-//      by the LS_FAST_SPGEMM switch — here the binning is automatic).
+//   cfg0: ub <=   48 —  8 lanes/row, 32 rows/block,   64-slot table
+//   cfg1: ub <=  128 — 16 lanes/row, 16 rows/block,  256-slot table
+//   cfg2: ub <= 1024 —  1 wave/row,   2 rows/block, 2048-slot table
+//   cfg3: ub <= 4096 —  1 block/row,               8192-slot table
+//   global bin (ub > 4096): per-row power-of-two table in HBM workspace,
+//     compacted unsorted (host sorts those rows; they are rare).
 //
-// Column ids are hashed as uint32 (requires B.cols < 2^32 — asserted on
-// the Python side).  Tables are linear-probed, empty slot = 0xFFFFFFFF.
-// Output columns are left UNSORTED within a row; the Python layer runs one
-// fused canonical sort when needed.
+// Column ids hash as uint32 (B.cols < 2^32, asserted in Python); tables
+// are linear-probed, empty slot = 0xFFFFFFFF (sorts to the end, so the
+// bitonic pass needs no size specialization).
 
 #include "common.h"
 
@@ -32,43 +33,32 @@ __device__ inline uint32_t hash1(uint32_t col, uint32_t mask) {
   return (col * 2654435761u) & mask;
 }
 
-// Insert col into an LDS/global table region; returns slot index or
-// 0xFFFFFFFF if it was already present.
-__device__ inline uint32_t tbl_insert(uint32_t* tbl, uint32_t mask,
-                                      uint32_t col) {
+// Insert col; returns true if newly inserted (symbolic counting).
+__device__ inline bool tbl_insert(uint32_t* tbl, uint32_t mask,
+                                  uint32_t col) {
   uint32_t h = hash1(col, mask);
   while (true) {
     uint32_t cur = tbl[h];
-    if (cur == col) return EMPTY;
+    if (cur == col) return false;
     if (cur == EMPTY) {
       uint32_t prev = atomicCAS(&tbl[h], EMPTY, col);
-      if (prev == EMPTY) return h;
-      if (prev == col) return EMPTY;
+      if (prev == EMPTY) return true;
+      if (prev == col) return false;
     }
     h = (h + 1) & mask;
   }
 }
 
-// Find the slot of an existing col (numeric accumulate path).
+// Find or claim the slot of col (numeric accumulate path).
 __device__ inline uint32_t tbl_find_or_insert(uint32_t* tbl, uint32_t mask,
-                                              uint32_t col, bool* fresh) {
+                                              uint32_t col) {
   uint32_t h = hash1(col, mask);
   while (true) {
     uint32_t cur = tbl[h];
-    if (cur == col) {
-      *fresh = false;
-      return h;
-    }
+    if (cur == col) return h;
     if (cur == EMPTY) {
       uint32_t prev = atomicCAS(&tbl[h], EMPTY, col);
-      if (prev == EMPTY) {
-        *fresh = true;
-        return h;
-      }
-      if (prev == col) {
-        *fresh = false;
-        return h;
-      }
+      if (prev == EMPTY || prev == col) return h;
     }
     h = (h + 1) & mask;
   }
@@ -85,8 +75,9 @@ __device__ inline void atomic_add_val(Cplx<T>* p, Cplx<T> v) {
 // ---------------------------------------------------------------------------
 // phase 0: upper bound per row
 // ---------------------------------------------------------------------------
+template <typename I>
 __global__ __launch_bounds__(LS_THREADS) void row_ub_kernel(
-    const i64* __restrict__ A_indptr, const i64* __restrict__ A_indices,
+    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
     const i64* __restrict__ B_indptr, i64* __restrict__ ub, i64 n_rows) {
   constexpr int W = 4;
   constexpr int RPB = LS_THREADS / W;
@@ -96,7 +87,7 @@ __global__ __launch_bounds__(LS_THREADS) void row_ub_kernel(
   for (i64 row = (i64)blockIdx.x * RPB + group; row < n_rows; row += stride) {
     i64 acc = 0;
     for (i64 jp = A_indptr[row] + lane; jp < A_indptr[row + 1]; jp += W) {
-      const i64 k = A_indices[jp];
+      const i64 k = (i64)A_indices[jp];
       acc += B_indptr[k + 1] - B_indptr[k];
     }
     acc = group_reduce_sum<i64, W>(acc);
@@ -107,13 +98,14 @@ __global__ __launch_bounds__(LS_THREADS) void row_ub_kernel(
 // ---------------------------------------------------------------------------
 // LDS-hash symbolic/numeric, parameterized on (BLOCK, ROWS_PER_BLOCK, TBL).
 // One row per thread-group (W = BLOCK/ROWS threads); exactly one row per
-// group per launch index (no grid-stride) so __syncthreads() is uniform.
+// group per launch index (no grid-stride) and compile-time loop bounds, so
+// every __syncthreads() is uniform across the block.
 // ---------------------------------------------------------------------------
-template <int BLOCK, int ROWS, int TBL>
+template <typename I, int BLOCK, int ROWS, int TBL>
 __global__ __launch_bounds__(BLOCK) void spgemm_symbolic_lds_kernel(
     const i64* __restrict__ rows_list, i64 n_bin_rows,
-    const i64* __restrict__ A_indptr, const i64* __restrict__ A_indices,
-    const i64* __restrict__ B_indptr, const i64* __restrict__ B_indices,
+    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
+    const i64* __restrict__ B_indptr, const I* __restrict__ B_indices,
     i64* __restrict__ row_nnz) {
   constexpr int W = BLOCK / ROWS;
   __shared__ uint32_t tbl[ROWS][TBL];
@@ -128,10 +120,9 @@ __global__ __launch_bounds__(BLOCK) void spgemm_symbolic_lds_kernel(
     const i64 row = rows_list[idx];
     int local = 0;
     for (i64 jp = A_indptr[row] + lane; jp < A_indptr[row + 1]; jp += W) {
-      const i64 k = A_indices[jp];
+      const i64 k = (i64)A_indices[jp];
       for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-        if (tbl_insert(tbl[group], TBL - 1,
-                       (uint32_t)B_indices[kp]) != EMPTY)
+        if (tbl_insert(tbl[group], TBL - 1, (uint32_t)B_indices[kp]))
           ++local;
       }
     }
@@ -141,18 +132,17 @@ __global__ __launch_bounds__(BLOCK) void spgemm_symbolic_lds_kernel(
   if (idx < n_bin_rows && lane == 0) row_nnz[rows_list[idx]] = cnt[group];
 }
 
-template <typename T, int BLOCK, int ROWS, int TBL>
+template <typename T, typename I, int BLOCK, int ROWS, int TBL>
 __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
     const i64* __restrict__ rows_list, i64 n_bin_rows,
-    const i64* __restrict__ A_indptr, const i64* __restrict__ A_indices,
+    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
     const T* __restrict__ A_vals, const i64* __restrict__ B_indptr,
-    const i64* __restrict__ B_indices, const T* __restrict__ B_vals,
-    const i64* __restrict__ C_indptr, i64* __restrict__ C_indices,
+    const I* __restrict__ B_indices, const T* __restrict__ B_vals,
+    const i64* __restrict__ C_indptr, I* __restrict__ C_indices,
     T* __restrict__ C_vals) {
   constexpr int W = BLOCK / ROWS;
   __shared__ uint32_t keys[ROWS][TBL];
   __shared__ T vals[ROWS][TBL];
-  __shared__ int cursor[ROWS];
   const int group = threadIdx.x / W;
   const int lane = threadIdx.x % W;
   const i64 idx = (i64)blockIdx.x * ROWS + group;
@@ -162,32 +152,52 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
     keys[group][t] = EMPTY;
     vals[group][t] = ls_zero<T>();
   }
-  if (lane == 0) cursor[group] = 0;
   __syncthreads();
   if (idx < n_bin_rows) {
     const i64 row = rows_list[idx];
     for (i64 jp = A_indptr[row] + lane; jp < A_indptr[row + 1]; jp += W) {
-      const i64 k = A_indices[jp];
+      const i64 k = (i64)A_indices[jp];
       const T a = A_vals[jp];
       for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-        bool fresh;
         const uint32_t slot = tbl_find_or_insert(
-            keys[group], TBL - 1, (uint32_t)B_indices[kp], &fresh);
+            keys[group], TBL - 1, (uint32_t)B_indices[kp]);
         atomic_add_val(&vals[group][slot], a * B_vals[kp]);
       }
     }
   }
   __syncthreads();
+  // in-place bitonic sort of (keys, vals) over the whole table; EMPTY
+  // (0xFFFFFFFF) sorts to the end, so the live entries end up as a sorted
+  // prefix.  Compile-time bounds -> uniform barriers.
+#pragma unroll 1
+  for (int k = 2; k <= TBL; k <<= 1) {
+#pragma unroll 1
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      for (int t = threadIdx.x % W; t < TBL; t += W) {
+        const int tj = t ^ j;
+        if (tj > t) {
+          const bool up = ((t & k) == 0);
+          const uint32_t a = keys[group][t];
+          const uint32_t b = keys[group][tj];
+          if ((a > b) == up) {
+            keys[group][t] = b;
+            keys[group][tj] = a;
+            const T tv = vals[group][t];
+            vals[group][t] = vals[group][tj];
+            vals[group][tj] = tv;
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
   if (idx < n_bin_rows) {
     const i64 row = rows_list[idx];
     const i64 base = C_indptr[row];
-    for (int t = lane; t < TBL; t += W) {
-      const uint32_t c = keys[group][t];
-      if (c != EMPTY) {
-        const int at = atomicAdd(&cursor[group], 1);
-        C_indices[base + at] = (i64)c;
-        C_vals[base + at] = vals[group][t];
-      }
+    const i64 cnt = C_indptr[row + 1] - base;
+    for (i64 t = lane; t < cnt; t += W) {
+      C_indices[base + t] = (I)keys[group][t];
+      C_vals[base + t] = vals[group][t];
     }
   }
 }
@@ -195,12 +205,14 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
 // ---------------------------------------------------------------------------
 // Global-memory hash bin (ub > LDS capacity): one 256-thread block per row,
 // per-row power-of-two table region in workspace (keys pre-filled 0xFF,
-// vals pre-zeroed by the host).
+// vals pre-zeroed by the host).  Output UNSORTED; host post-sorts these
+// rare rows.
 // ---------------------------------------------------------------------------
+template <typename I>
 __global__ __launch_bounds__(LS_THREADS) void spgemm_symbolic_global_kernel(
     const i64* __restrict__ rows_list, i64 n_bin_rows,
-    const i64* __restrict__ A_indptr, const i64* __restrict__ A_indices,
-    const i64* __restrict__ B_indptr, const i64* __restrict__ B_indices,
+    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
+    const i64* __restrict__ B_indptr, const I* __restrict__ B_indices,
     uint32_t* __restrict__ tbl_ws, const i64* __restrict__ tbl_off,
     const i64* __restrict__ tbl_size, i64* __restrict__ row_nnz) {
   const i64 idx = blockIdx.x;
@@ -214,9 +226,9 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_symbolic_global_kernel(
   int local = 0;
   for (i64 jp = A_indptr[row] + threadIdx.x; jp < A_indptr[row + 1];
        jp += blockDim.x) {
-    const i64 k = A_indices[jp];
+    const i64 k = (i64)A_indices[jp];
     for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-      if (tbl_insert(tbl, mask, (uint32_t)B_indices[kp]) != EMPTY) ++local;
+      if (tbl_insert(tbl, mask, (uint32_t)B_indices[kp])) ++local;
     }
   }
   atomicAdd(&cnt, local);
@@ -224,15 +236,15 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_symbolic_global_kernel(
   if (threadIdx.x == 0) row_nnz[row] = cnt;
 }
 
-template <typename T>
+template <typename T, typename I>
 __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_kernel(
     const i64* __restrict__ rows_list, i64 n_bin_rows,
-    const i64* __restrict__ A_indptr, const i64* __restrict__ A_indices,
+    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
     const T* __restrict__ A_vals, const i64* __restrict__ B_indptr,
-    const i64* __restrict__ B_indices, const T* __restrict__ B_vals,
+    const I* __restrict__ B_indices, const T* __restrict__ B_vals,
     uint32_t* __restrict__ key_ws, T* __restrict__ val_ws,
     const i64* __restrict__ tbl_off, const i64* __restrict__ tbl_size,
-    const i64* __restrict__ C_indptr, i64* __restrict__ C_indices,
+    const i64* __restrict__ C_indptr, I* __restrict__ C_indices,
     T* __restrict__ C_vals) {
   const i64 idx = blockIdx.x;
   if (idx >= n_bin_rows) return;
@@ -245,13 +257,11 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_kernel(
   __syncthreads();
   for (i64 jp = A_indptr[row] + threadIdx.x; jp < A_indptr[row + 1];
        jp += blockDim.x) {
-    const i64 k = A_indices[jp];
+    const i64 k = (i64)A_indices[jp];
     const T a = A_vals[jp];
     for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-      bool fresh;
       const uint32_t slot = tbl_find_or_insert(
-          keys, mask, (uint32_t)B_indices[kp], &fresh);
-      // val_ws is pre-zeroed by the host, so no init race exists here.
+          keys, mask, (uint32_t)B_indices[kp]);
       atomic_add_val(&vals[slot], a * B_vals[kp]);
     }
   }
@@ -261,8 +271,71 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_kernel(
     const uint32_t c = keys[t];
     if (c != EMPTY) {
       const int at = atomicAdd(&cursor, 1);
-      C_indices[base + at] = (i64)c;
+      C_indices[base + at] = (I)c;
       C_vals[base + at] = vals[t];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Launch helpers
+// ---------------------------------------------------------------------------
+template <typename I>
+void symbolic_lds_dispatch(int cfg, const i64* rl, i64 n, const i64* ap,
+                           const I* ai, const i64* bp, const I* bi,
+                           i64* rn, hipStream_t s) {
+  switch (cfg) {
+    case 0:
+      hipLaunchKernelGGL((spgemm_symbolic_lds_kernel<I, 256, 32, 64>),
+                         dim3((n + 31) / 32), dim3(256), 0, s, rl, n, ap,
+                         ai, bp, bi, rn);
+      break;
+    case 1:
+      hipLaunchKernelGGL((spgemm_symbolic_lds_kernel<I, 256, 16, 256>),
+                         dim3((n + 15) / 16), dim3(256), 0, s, rl, n, ap,
+                         ai, bp, bi, rn);
+      break;
+    case 2:
+      hipLaunchKernelGGL((spgemm_symbolic_lds_kernel<I, 128, 2, 2048>),
+                         dim3((n + 1) / 2), dim3(128), 0, s, rl, n, ap, ai,
+                         bp, bi, rn);
+      break;
+    default:
+      hipLaunchKernelGGL((spgemm_symbolic_lds_kernel<I, 256, 1, 8192>),
+                         dim3(n), dim3(256), 0, s, rl, n, ap, ai, bp, bi,
+                         rn);
+  }
+}
+
+template <typename T, typename I>
+void numeric_lds_dispatch(int cfg, const i64* rl, i64 n, const i64* ap,
+                          const I* ai, const T* av, const i64* bp,
+                          const I* bi, const T* bv, const i64* cp, I* ci,
+                          T* cv, hipStream_t s) {
+  switch (cfg) {
+    case 0:
+      hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, I, 256, 32, 64>),
+                         dim3((n + 31) / 32), dim3(256), 0, s, rl, n, ap,
+                         ai, av, bp, bi, bv, cp, ci, cv);
+      break;
+    case 1:
+      hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, I, 256, 16, 256>),
+                         dim3((n + 15) / 16), dim3(256), 0, s, rl, n, ap,
+                         ai, av, bp, bi, bv, cp, ci, cv);
+      break;
+    case 2:
+      hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, I, 128, 2, 2048>),
+                         dim3((n + 1) / 2), dim3(128), 0, s, rl, n, ap, ai,
+                         av, bp, bi, bv, cp, ci, cv);
+      break;
+    default: {
+      // 16-byte values (c128) at TBL=8192 exceed 160 KiB LDS; use 4096
+      // slots there (bin cutoff is ub <= 4096: the table can be exactly
+      // full — linear probing still terminates).
+      constexpr int TBL3 = (sizeof(T) > 8) ? 4096 : 8192;
+      hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, I, 256, 1, TBL3>),
+                         dim3(n), dim3(256), 0, s, rl, n, ap, ai, av, bp,
+                         bi, bv, cp, ci, cv);
     }
   }
 }
@@ -274,71 +347,32 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_kernel(
 // ---------------------------------------------------------------------------
 void ls_spgemm_row_ub(uintptr_t A_indptr, uintptr_t A_indices,
                       uintptr_t B_indptr, uintptr_t ub, i64 n_rows,
-                      uintptr_t stream) {
+                      int idx_dtype, uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   int grid = grid_1d(n_rows, LS_THREADS / 4, 8192);
-  hipLaunchKernelGGL(row_ub_kernel, dim3(grid), dim3(LS_THREADS), 0, s,
-                     reinterpret_cast<const i64*>(A_indptr),
-                     reinterpret_cast<const i64*>(A_indices),
-                     reinterpret_cast<const i64*>(B_indptr),
-                     reinterpret_cast<i64*>(ub), n_rows);
+  DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
+      (row_ub_kernel<idx_t>), dim3(grid), dim3(LS_THREADS), 0, s,
+      reinterpret_cast<const i64*>(A_indptr),
+      reinterpret_cast<const idx_t*>(A_indices),
+      reinterpret_cast<const i64*>(B_indptr),
+      reinterpret_cast<i64*>(ub), n_rows));
   ls_check(hipGetLastError(), "spgemm_row_ub");
 }
 
 void ls_spgemm_symbolic_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
                             uintptr_t A_indptr, uintptr_t A_indices,
                             uintptr_t B_indptr, uintptr_t B_indices,
-                            uintptr_t row_nnz, uintptr_t stream) {
+                            uintptr_t row_nnz, int idx_dtype,
+                            uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
-  auto rl = reinterpret_cast<const i64*>(rows_list);
-  auto ap = reinterpret_cast<const i64*>(A_indptr);
-  auto ai = reinterpret_cast<const i64*>(A_indices);
-  auto bp = reinterpret_cast<const i64*>(B_indptr);
-  auto bi = reinterpret_cast<const i64*>(B_indices);
-  auto rn = reinterpret_cast<i64*>(row_nnz);
-  if (cfg == 0) {
-    const i64 grid = (n_bin_rows + 15) / 16;
-    hipLaunchKernelGGL((spgemm_symbolic_lds_kernel<256, 16, 256>),
-                       dim3(grid), dim3(256), 0, s, rl, n_bin_rows, ap, ai,
-                       bp, bi, rn);
-  } else if (cfg == 1) {
-    const i64 grid = (n_bin_rows + 1) / 2;
-    hipLaunchKernelGGL((spgemm_symbolic_lds_kernel<128, 2, 2048>),
-                       dim3(grid), dim3(128), 0, s, rl, n_bin_rows, ap, ai,
-                       bp, bi, rn);
-  } else {
-    hipLaunchKernelGGL((spgemm_symbolic_lds_kernel<256, 1, 8192>),
-                       dim3(n_bin_rows), dim3(256), 0, s, rl, n_bin_rows,
-                       ap, ai, bp, bi, rn);
-  }
+  DISPATCH_IDX_T(idx_dtype, symbolic_lds_dispatch<idx_t>(
+      cfg, reinterpret_cast<const i64*>(rows_list), n_bin_rows,
+      reinterpret_cast<const i64*>(A_indptr),
+      reinterpret_cast<const idx_t*>(A_indices),
+      reinterpret_cast<const i64*>(B_indptr),
+      reinterpret_cast<const idx_t*>(B_indices),
+      reinterpret_cast<i64*>(row_nnz), s));
   ls_check(hipGetLastError(), "spgemm_symbolic_lds");
-}
-
-template <typename T>
-static void numeric_lds_dispatch(int cfg, const i64* rl, i64 n,
-                                 const i64* ap, const i64* ai, const T* av,
-                                 const i64* bp, const i64* bi, const T* bv,
-                                 const i64* cp, i64* ci, T* cv,
-                                 hipStream_t s) {
-  if (cfg == 0) {
-    const i64 grid = (n + 15) / 16;
-    hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, 256, 16, 256>),
-                       dim3(grid), dim3(256), 0, s, rl, n, ap, ai, av, bp,
-                       bi, bv, cp, ci, cv);
-  } else if (cfg == 1) {
-    const i64 grid = (n + 1) / 2;
-    hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, 128, 2, 2048>),
-                       dim3(grid), dim3(128), 0, s, rl, n, ap, ai, av, bp,
-                       bi, bv, cp, ci, cv);
-  } else {
-    // 16-byte values (c128) at TBL=8192 would need 160 KiB + change of
-    // LDS; drop to 4096 slots there (bin cutoff is ub <= 4096, so the
-    // table can be exactly full — linear probing still terminates).
-    constexpr int TBL2 = (sizeof(T) > 8) ? 4096 : 8192;
-    hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, 256, 1, TBL2>),
-                       dim3(n), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi,
-                       bv, cp, ci, cv);
-  }
 }
 
 void ls_spgemm_numeric_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
@@ -346,19 +380,21 @@ void ls_spgemm_numeric_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
                            uintptr_t A_vals, uintptr_t B_indptr,
                            uintptr_t B_indices, uintptr_t B_vals,
                            uintptr_t C_indptr, uintptr_t C_indices,
-                           uintptr_t C_vals, int dtype, uintptr_t stream) {
+                           uintptr_t C_vals, int dtype, int idx_dtype,
+                           uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
-  DISPATCH_VAL_T(dtype, numeric_lds_dispatch<val_t>(
-      cfg, reinterpret_cast<const i64*>(rows_list), n_bin_rows,
-      reinterpret_cast<const i64*>(A_indptr),
-      reinterpret_cast<const i64*>(A_indices),
-      reinterpret_cast<const val_t*>(A_vals),
-      reinterpret_cast<const i64*>(B_indptr),
-      reinterpret_cast<const i64*>(B_indices),
-      reinterpret_cast<const val_t*>(B_vals),
-      reinterpret_cast<const i64*>(C_indptr),
-      reinterpret_cast<i64*>(C_indices), reinterpret_cast<val_t*>(C_vals),
-      s));
+  DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
+      numeric_lds_dispatch<val_t, idx_t>(
+          cfg, reinterpret_cast<const i64*>(rows_list), n_bin_rows,
+          reinterpret_cast<const i64*>(A_indptr),
+          reinterpret_cast<const idx_t*>(A_indices),
+          reinterpret_cast<const val_t*>(A_vals),
+          reinterpret_cast<const i64*>(B_indptr),
+          reinterpret_cast<const idx_t*>(B_indices),
+          reinterpret_cast<const val_t*>(B_vals),
+          reinterpret_cast<const i64*>(C_indptr),
+          reinterpret_cast<idx_t*>(C_indices),
+          reinterpret_cast<val_t*>(C_vals), s))));
   ls_check(hipGetLastError(), "spgemm_numeric_lds");
 }
 
@@ -367,19 +403,19 @@ void ls_spgemm_symbolic_global(uintptr_t rows_list, i64 n_bin_rows,
                                uintptr_t B_indptr, uintptr_t B_indices,
                                uintptr_t tbl_ws, uintptr_t tbl_off,
                                uintptr_t tbl_size, uintptr_t row_nnz,
-                               uintptr_t stream) {
+                               int idx_dtype, uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
-  hipLaunchKernelGGL(spgemm_symbolic_global_kernel, dim3(n_bin_rows),
-                     dim3(LS_THREADS), 0, s,
-                     reinterpret_cast<const i64*>(rows_list), n_bin_rows,
-                     reinterpret_cast<const i64*>(A_indptr),
-                     reinterpret_cast<const i64*>(A_indices),
-                     reinterpret_cast<const i64*>(B_indptr),
-                     reinterpret_cast<const i64*>(B_indices),
-                     reinterpret_cast<uint32_t*>(tbl_ws),
-                     reinterpret_cast<const i64*>(tbl_off),
-                     reinterpret_cast<const i64*>(tbl_size),
-                     reinterpret_cast<i64*>(row_nnz));
+  DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
+      (spgemm_symbolic_global_kernel<idx_t>), dim3(n_bin_rows),
+      dim3(LS_THREADS), 0, s, reinterpret_cast<const i64*>(rows_list),
+      n_bin_rows, reinterpret_cast<const i64*>(A_indptr),
+      reinterpret_cast<const idx_t*>(A_indices),
+      reinterpret_cast<const i64*>(B_indptr),
+      reinterpret_cast<const idx_t*>(B_indices),
+      reinterpret_cast<uint32_t*>(tbl_ws),
+      reinterpret_cast<const i64*>(tbl_off),
+      reinterpret_cast<const i64*>(tbl_size),
+      reinterpret_cast<i64*>(row_nnz)));
   ls_check(hipGetLastError(), "spgemm_symbolic_global");
 }
 
@@ -390,23 +426,24 @@ void ls_spgemm_numeric_global(uintptr_t rows_list, i64 n_bin_rows,
                               uintptr_t key_ws, uintptr_t val_ws,
                               uintptr_t tbl_off, uintptr_t tbl_size,
                               uintptr_t C_indptr, uintptr_t C_indices,
-                              uintptr_t C_vals, int dtype, uintptr_t stream) {
+                              uintptr_t C_vals, int dtype, int idx_dtype,
+                              uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
-  DISPATCH_VAL_T(dtype, hipLaunchKernelGGL(
-      (spgemm_numeric_global_kernel<val_t>), dim3(n_bin_rows),
+  DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
+      (spgemm_numeric_global_kernel<val_t, idx_t>), dim3(n_bin_rows),
       dim3(LS_THREADS), 0, s, reinterpret_cast<const i64*>(rows_list),
       n_bin_rows, reinterpret_cast<const i64*>(A_indptr),
-      reinterpret_cast<const i64*>(A_indices),
+      reinterpret_cast<const idx_t*>(A_indices),
       reinterpret_cast<const val_t*>(A_vals),
       reinterpret_cast<const i64*>(B_indptr),
-      reinterpret_cast<const i64*>(B_indices),
+      reinterpret_cast<const idx_t*>(B_indices),
       reinterpret_cast<const val_t*>(B_vals),
       reinterpret_cast<uint32_t*>(key_ws),
       reinterpret_cast<val_t*>(val_ws),
       reinterpret_cast<const i64*>(tbl_off),
       reinterpret_cast<const i64*>(tbl_size),
       reinterpret_cast<const i64*>(C_indptr),
-      reinterpret_cast<i64*>(C_indices),
-      reinterpret_cast<val_t*>(C_vals)));
+      reinterpret_cast<idx_t*>(C_indices),
+      reinterpret_cast<val_t*>(C_vals))));
   ls_check(hipGetLastError(), "spgemm_numeric_global");
 }

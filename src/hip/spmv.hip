@@ -8,18 +8,30 @@
 // Strategy: "vector CSR" — a power-of-two sub-wave of W lanes per row,
 // W picked by the host from the mean row length.  Within a row the W lanes
 // read vals/indices coalesced; sub-wave reduction via __shfl_down (64-wide
-// waves, cdna_hip_programming.md §1).  fp64 CSR SpMV is gather/memory-bound
-// (~12-16 B/nnz), so the job is purely to keep the load pipes full:
-// grid-stride with a capped grid (G11), no LDS staging (x windows L2/L3-fit
-// for banded matrices; 256 MiB Infinity Cache holds x up to 32M fp64).
+// waves, cdna_hip_programming.md §1).  fp64 CSR SpMV is memory-bound
+// (12-16 B/nnz), so the design minimizes streamed bytes:
+//   - int32 column indices when the matrix allows (DISPATCH_IDX_T)
+//   - non-temporal loads on vals/indices (streamed exactly once per call;
+//     x stays cache-friendly: L3 holds x up to 32M fp64)
+//   - grid-stride with a capped grid (G11), no LDS staging.
 
 #include "common.h"
 
 namespace {
 
-template <typename T, int W>
+template <typename T>
+__device__ inline T nt_load(const T* p) {
+  return __builtin_nontemporal_load(p);
+}
+template <typename T>
+__device__ inline Cplx<T> nt_load(const Cplx<T>* p) {
+  return {__builtin_nontemporal_load(&p->re),
+          __builtin_nontemporal_load(&p->im)};
+}
+
+template <typename T, typename I, int W>
 __global__ __launch_bounds__(LS_THREADS) void spmv_vector_kernel(
-    const i64* __restrict__ indptr, const i64* __restrict__ indices,
+    const i64* __restrict__ indptr, const I* __restrict__ indices,
     const T* __restrict__ vals, const T* __restrict__ x, T* __restrict__ y,
     i64 n_rows, int accumulate) {
   constexpr int ROWS_PER_BLOCK = LS_THREADS / W;
@@ -32,7 +44,7 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_vector_kernel(
     const i64 e = indptr[row + 1];
     T acc = ls_zero<T>();
     for (i64 jp = s + lane; jp < e; jp += W) {
-      acc += vals[jp] * x[indices[jp]];
+      acc += nt_load(vals + jp) * x[nt_load(indices + jp)];
     }
     acc = group_reduce_sum<T, W>(acc);
     if (lane == 0) {
@@ -44,8 +56,8 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_vector_kernel(
   }
 }
 
-template <typename T>
-void spmv_launch(const i64* indptr, const i64* indices, const T* vals,
+template <typename T, typename I>
+void spmv_launch(const i64* indptr, const I* indices, const T* vals,
                  const T* x, T* y, i64 n_rows, i64 nnz, bool accumulate,
                  hipStream_t stream) {
   const double mean = n_rows > 0 ? (double)nnz / (double)n_rows : 0.0;
@@ -57,7 +69,7 @@ void spmv_launch(const i64* indptr, const i64* indices, const T* vals,
     constexpr int WS = decltype(wtag)::value;
     constexpr int RPB = LS_THREADS / WS;
     int grid = grid_1d(n_rows, RPB, 8192);
-    hipLaunchKernelGGL((spmv_vector_kernel<T, WS>), dim3(grid),
+    hipLaunchKernelGGL((spmv_vector_kernel<T, I, WS>), dim3(grid),
                        dim3(LS_THREADS), 0, stream, indptr, indices, vals, x,
                        y, n_rows, accumulate ? 1 : 0);
   };
@@ -77,12 +89,13 @@ void spmv_launch(const i64* indptr, const i64* indices, const T* vals,
 
 void ls_spmv(uintptr_t indptr, uintptr_t indices, uintptr_t vals,
              uintptr_t x, uintptr_t y, i64 n_rows, i64 nnz, int dtype,
-             bool accumulate, uintptr_t stream) {
+             int idx_dtype, bool accumulate, uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
-  DISPATCH_VAL_T(dtype, spmv_launch<val_t>(
-      reinterpret_cast<const i64*>(indptr),
-      reinterpret_cast<const i64*>(indices),
-      reinterpret_cast<const val_t*>(vals),
-      reinterpret_cast<const val_t*>(x), reinterpret_cast<val_t*>(y),
-      n_rows, nnz, accumulate, s));
+  DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
+      spmv_launch<val_t, idx_t>(
+          reinterpret_cast<const i64*>(indptr),
+          reinterpret_cast<const idx_t*>(indices),
+          reinterpret_cast<const val_t*>(vals),
+          reinterpret_cast<const val_t*>(x), reinterpret_cast<val_t*>(y),
+          n_rows, nnz, accumulate, s))));
 }
