@@ -1,0 +1,77 @@
+#!/usr/bin/env python3
+# SPDX-License-Identifier: Apache-2.0
+"""Within-process interleaved A/B sweep of SpMV variants (guide §5.4
+rule 24: interleave variants in ONE process, report median).
+
+Sweeps {W} x {nt} x {index width} on the 5-pt Poisson operator and a
+banded matrix.  Run on the GPU box:
+    python benchmarks/spmv_sweep.py [--nx 4096] [--rounds 7]
+"""
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+import torch
+
+from legate_sparse import ops
+from legate_sparse.gallery import poisson_2d
+from legate_sparse.runtime import runtime
+
+
+def time_variant(A, x, y, w, nt, iters=30):
+    idp, idx, dat = A._indptr, A._indices, A._data
+    for _ in range(3):
+        ops.spmv(idp, idx, dat, x, y, w_override=w, nt=nt)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        ops.spmv(idp, idx, dat, x, y, w_override=w, nt=nt)
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--nx", type=int, default=4096)
+    ap.add_argument("--rounds", type=int, default=7)
+    ap.add_argument("--iters", type=int, default=30)
+    args = ap.parse_args()
+    assert torch.cuda.is_available()
+
+    A = poisson_2d(args.nx, args.nx)
+    n = A.shape[0]
+    nnz = A.nnz
+    x = torch.rand(n, dtype=torch.float64, device="cuda")
+    y = torch.empty_like(x)
+    # int64-index clone (bypasses _init_local's int32 cast)
+    import copy
+    A64 = copy.copy(A)
+    A64._indices = A._indices.to(torch.int64).contiguous()
+
+    variants = []
+    for w in (2, 4, 8):
+        for nt in (False, True):
+            variants.append(("i32", A, w, nt))
+    variants.append(("i64", A64, 4, False))
+    variants.append(("i64", A64, 4, True))
+
+    results = {v[:1] + v[2:]: [] for v in variants}
+    for r in range(args.rounds):
+        for tag, mat, w, nt in variants:
+            dt = time_variant(mat, x, y, w, nt, args.iters)
+            results[(tag, w, nt)].append(dt)
+    print(f"# nx={args.nx} rows={n} nnz={nnz}")
+    for (tag, w, nt), ts in sorted(results.items()):
+        med = np.median(ts) * 1e3
+        mn = min(ts) * 1e3
+        gf = 2 * nnz / (np.median(ts)) / 1e9
+        print(f"idx={tag} W={w} nt={int(nt)}: median {med:.4f} ms "
+              f"(min {mn:.4f})  {gf:7.1f} GFLOP/s")
+
+
+if __name__ == "__main__":
+    main()

@@ -29,7 +29,7 @@ __device__ inline Cplx<T> nt_load(const Cplx<T>* p) {
           __builtin_nontemporal_load(&p->im)};
 }
 
-template <typename T, typename I, int W>
+template <typename T, typename I, int W, bool NT>
 __global__ __launch_bounds__(LS_THREADS) void spmv_vector_kernel(
     const i64* __restrict__ indptr, const I* __restrict__ indices,
     const T* __restrict__ vals, const T* __restrict__ x, T* __restrict__ y,
@@ -44,7 +44,10 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_vector_kernel(
     const i64 e = indptr[row + 1];
     T acc = ls_zero<T>();
     for (i64 jp = s + lane; jp < e; jp += W) {
-      acc += nt_load(vals + jp) * x[nt_load(indices + jp)];
+      if constexpr (NT)
+        acc += nt_load(vals + jp) * x[nt_load(indices + jp)];
+      else
+        acc += vals[jp] * x[indices[jp]];
     }
     acc = group_reduce_sum<T, W>(acc);
     if (lane == 0) {
@@ -59,19 +62,25 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_vector_kernel(
 template <typename T, typename I>
 void spmv_launch(const i64* indptr, const I* indices, const T* vals,
                  const T* x, T* y, i64 n_rows, i64 nnz, bool accumulate,
-                 hipStream_t stream) {
+                 int w_override, bool nt, hipStream_t stream) {
   const double mean = n_rows > 0 ? (double)nnz / (double)n_rows : 0.0;
   // W = largest power of two <= mean (clamped): short rows get small
   // groups (less reduction overhead), long rows get coalesced width.
   int W = 1;
   while (W < 64 && (double)(W * 2) <= mean) W *= 2;
+  if (w_override > 0) W = w_override;
   auto launch = [&](auto wtag) {
     constexpr int WS = decltype(wtag)::value;
     constexpr int RPB = LS_THREADS / WS;
     int grid = grid_1d(n_rows, RPB, 8192);
-    hipLaunchKernelGGL((spmv_vector_kernel<T, I, WS>), dim3(grid),
-                       dim3(LS_THREADS), 0, stream, indptr, indices, vals, x,
-                       y, n_rows, accumulate ? 1 : 0);
+    if (nt)
+      hipLaunchKernelGGL((spmv_vector_kernel<T, I, WS, true>), dim3(grid),
+                         dim3(LS_THREADS), 0, stream, indptr, indices, vals,
+                         x, y, n_rows, accumulate ? 1 : 0);
+    else
+      hipLaunchKernelGGL((spmv_vector_kernel<T, I, WS, false>), dim3(grid),
+                         dim3(LS_THREADS), 0, stream, indptr, indices, vals,
+                         x, y, n_rows, accumulate ? 1 : 0);
   };
   switch (W) {
     case 1: launch(std::integral_constant<int, 1>{}); break;
@@ -89,7 +98,8 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
 
 void ls_spmv(uintptr_t indptr, uintptr_t indices, uintptr_t vals,
              uintptr_t x, uintptr_t y, i64 n_rows, i64 nnz, int dtype,
-             int idx_dtype, bool accumulate, uintptr_t stream) {
+             int idx_dtype, bool accumulate, uintptr_t stream,
+             int w_override, bool nt) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
       spmv_launch<val_t, idx_t>(
@@ -97,5 +107,5 @@ void ls_spmv(uintptr_t indptr, uintptr_t indices, uintptr_t vals,
           reinterpret_cast<const idx_t*>(indices),
           reinterpret_cast<const val_t*>(vals),
           reinterpret_cast<const val_t*>(x), reinterpret_cast<val_t*>(y),
-          n_rows, nnz, accumulate, s))));
+          n_rows, nnz, accumulate, w_override, nt, s))));
 }
