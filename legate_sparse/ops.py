@@ -51,7 +51,7 @@ def _stream() -> int:
 def spmv(indptr: torch.Tensor, indices: torch.Tensor, vals: torch.Tensor,
          x: torch.Tensor, y: Optional[torch.Tensor] = None,
          accumulate: bool = False, w_override: int = 0,
-         nt: bool = True) -> torch.Tensor:
+         nt: bool = False) -> torch.Tensor:
     n_rows = indptr.numel() - 1
     if y is None:
         y = torch.empty(n_rows, dtype=vals.dtype, device=vals.device)

@@ -11,8 +11,9 @@
 // waves, cdna_hip_programming.md §1).  fp64 CSR SpMV is memory-bound
 // (12-16 B/nnz), so the design minimizes streamed bytes:
 //   - int32 column indices when the matrix allows (DISPATCH_IDX_T)
-//   - non-temporal loads on vals/indices (streamed exactly once per call;
-//     x stays cache-friendly: L3 holds x up to 32M fp64)
+//   - plain (cached) loads: a measured nt-load variant was 25-40%% SLOWER
+//     (profiles/spmv_sweep_r01.txt) — nt defeats the L2/L3 help on the
+//     vals/indices stream here, unlike the decode-weight case
 //   - grid-stride with a capped grid (G11), no LDS staging.
 
 #include "common.h"
@@ -64,10 +65,11 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
                  const T* x, T* y, i64 n_rows, i64 nnz, bool accumulate,
                  int w_override, bool nt, hipStream_t stream) {
   const double mean = n_rows > 0 ? (double)nnz / (double)n_rows : 0.0;
-  // W = largest power of two <= mean (clamped): short rows get small
-  // groups (less reduction overhead), long rows get coalesced width.
+  // W = largest power of two <= mean/2 (measured on MI355X: ~2 elements
+  // per lane beats 1; profiles/spmv_sweep_r01.txt — W=2 gives 504 GF/s on
+  // 5-pt Poisson fp64 vs 442 at W=4).
   int W = 1;
-  while (W < 64 && (double)(W * 2) <= mean) W *= 2;
+  while (W < 64 && (double)(W * 4) <= mean) W *= 2;
   if (w_override > 0) W = w_override;
   auto launch = [&](auto wtag) {
     constexpr int WS = decltype(wtag)::value;
