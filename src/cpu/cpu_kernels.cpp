@@ -31,7 +31,9 @@ using i64 = int64_t;
 template <typename T, typename I>
 static void spmv_impl(const i64* indptr, const I* indices, const T* vals,
                       const T* x, T* y, i64 n_rows, bool accumulate) {
-#pragma omp parallel for schedule(dynamic, 256)
+  // `if` clause: a parallel region on a tiny matrix costs more than the
+  // work itself (and spins cgroup CPU quota in containers)
+#pragma omp parallel for schedule(static) if (n_rows > 16384)
   for (i64 i = 0; i < n_rows; ++i) {
     T acc = accumulate ? y[i] : T(0);
     for (i64 jp = indptr[i]; jp < indptr[i + 1]; ++jp) {
@@ -47,7 +49,7 @@ static void spgemm_symbolic_impl(const i64* A_indptr, const I* A_indices,
                                  i64 n_rowsA, const i64* B_indptr,
                                  const I* B_indices, i64 n_colsB,
                                  i64* row_nnz) {
-#pragma omp parallel
+#pragma omp parallel if (n_rowsA > 4096)
   {
     std::vector<i64> marker(n_colsB, -1);
 #pragma omp for schedule(dynamic, 64)
@@ -78,7 +80,7 @@ static void spgemm_numeric_impl(const i64* A_indptr, const I* A_indices,
                                 const T* B_vals, i64 n_colsB,
                                 const i64* C_indptr, I* C_indices,
                                 T* C_vals) {
-#pragma omp parallel
+#pragma omp parallel if (n_rowsA > 4096)
   {
     std::vector<T> workspace(n_colsB, T(0));
     std::vector<i64> marker(n_colsB, -1);

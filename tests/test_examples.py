@@ -1,0 +1,81 @@
+# SPDX-License-Identifier: Apache-2.0
+"""The example drivers (the reference's headline workloads) run end-to-end
+and pass their own residual checks."""
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run(script, *args, timeout=300):
+    r = subprocess.run([sys.executable, os.path.join(REPO, "examples",
+                                                     script), *args],
+                       capture_output=True, text=True, timeout=timeout)
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    return r.stdout
+
+
+def test_pde_runs():
+    out = _run("pde.py", "-n", "64")
+    assert "PASS" in out
+
+
+def test_pde_throughput_mode():
+    out = _run("pde.py", "-n", "64", "--throughput", "-i", "20", "-w", "2")
+    assert "ms/iter" in out
+
+
+def test_gmg_runs():
+    out = _run("gmg.py", "-k", "5")
+    assert "PASS" in out
+
+
+def test_spmv_microbenchmark():
+    out = _run("spmv_microbenchmark.py", "--nmin", "4096", "--nmax", "8192",
+               "-i", "3", "-w", "1")
+    assert "ms/iter" in out or "GFLOP" in out
+
+
+def test_spgemm_microbenchmark():
+    out = _run("spgemm_microbenchmark.py", "-n", "4096", "-i", "2", "-w",
+               "1")
+    assert "ms/iter" in out
+
+
+def test_bench_contract():
+    """bench.py prints exactly one valid JSON line with required keys."""
+    import json
+    r = subprocess.run([sys.executable, os.path.join(REPO, "bench.py"),
+                        "--nx", "128", "--steps", "5", "--warmup", "1",
+                        "--spgemm-iters", "1"],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling",
+                "vs_baseline", "dtype", "data", "config"):
+        assert key in d, key
+    assert d["n_gpus"] == 1
+    assert d["scaling"] == "weak"
+
+
+@pytest.mark.gpu
+def test_pde_gpu():
+    import torch
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    out = _run("pde.py", "-n", "256")
+    assert "PASS" in out
+
+
+@pytest.mark.gpu
+def test_gmg_gpu():
+    import torch
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    out = _run("gmg.py", "-k", "7")
+    assert "PASS" in out
