@@ -22,14 +22,14 @@ from legate_sparse.gallery import poisson_2d
 from legate_sparse.runtime import runtime
 
 
-def time_variant(A, x, y, w, nt, iters=30):
+def time_variant(A, x, y, w, pair, swz, iters=30):
     idp, idx, dat = A._indptr, A._indices, A._data
     for _ in range(3):
-        ops.spmv(idp, idx, dat, x, y, w_override=w, nt=nt)
+        ops.spmv(idp, idx, dat, x, y, w_override=w, pair=pair, swz=swz)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        ops.spmv(idp, idx, dat, x, y, w_override=w, nt=nt)
+        ops.spmv(idp, idx, dat, x, y, w_override=w, pair=pair, swz=swz)
     torch.cuda.synchronize()
     return (time.perf_counter() - t0) / iters
 
@@ -53,23 +53,22 @@ def main():
     A64._indices = A._indices.to(torch.int64).contiguous()
 
     variants = []
-    for w in (2, 4, 8):
-        for nt in (False, True):
-            variants.append(("i32", A, w, nt))
-    variants.append(("i64", A64, 4, False))
-    variants.append(("i64", A64, 4, True))
+    for w in (1, 2, 4):
+        for pair in (0, 1):
+            for swz in (0, 1):
+                variants.append(("i32", A, w, pair, swz))
 
     results = {v[:1] + v[2:]: [] for v in variants}
     for r in range(args.rounds):
-        for tag, mat, w, nt in variants:
-            dt = time_variant(mat, x, y, w, nt, args.iters)
-            results[(tag, w, nt)].append(dt)
+        for tag, mat, w, pair, swz in variants:
+            dt = time_variant(mat, x, y, w, pair, swz, args.iters)
+            results[(tag, w, pair, swz)].append(dt)
     print(f"# nx={args.nx} rows={n} nnz={nnz}")
-    for (tag, w, nt), ts in sorted(results.items()):
+    for (tag, w, pair, swz), ts in sorted(results.items()):
         med = np.median(ts) * 1e3
         mn = min(ts) * 1e3
         gf = 2 * nnz / (np.median(ts)) / 1e9
-        print(f"idx={tag} W={w} nt={int(nt)}: median {med:.4f} ms "
+        print(f"idx={tag} W={w} pair={pair} swz={swz}: median {med:.4f} ms "
               f"(min {mn:.4f})  {gf:7.1f} GFLOP/s")
 
 

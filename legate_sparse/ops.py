@@ -51,7 +51,7 @@ def _stream() -> int:
 def spmv(indptr: torch.Tensor, indices: torch.Tensor, vals: torch.Tensor,
          x: torch.Tensor, y: Optional[torch.Tensor] = None,
          accumulate: bool = False, w_override: int = 0,
-         nt: bool = False) -> torch.Tensor:
+         nt: bool = False, pair: int = -1, swz: int = 0) -> torch.Tensor:
     n_rows = indptr.numel() - 1
     if y is None:
         y = torch.empty(n_rows, dtype=vals.dtype, device=vals.device)
@@ -61,7 +61,7 @@ def spmv(indptr: torch.Tensor, indices: torch.Tensor, vals: torch.Tensor,
         ext.spmv(indptr.data_ptr(), indices.data_ptr(), vals.data_ptr(),
                  x.data_ptr(), y.data_ptr(), n_rows, vals.numel(),
                  _code(vals), _icode(indices), accumulate, _stream(),
-                 int(w_override), bool(nt))
+                 int(w_override), bool(nt), int(pair), int(swz))
         return y
     if not vals.is_cuda and _cext.has_cpu():
         _cext.require_cpu().spmv(indptr.data_ptr(), indices.data_ptr(),

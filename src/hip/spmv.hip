@@ -5,16 +5,19 @@
 // (the reference computes this through cuSPARSE on a localized CSR,
 // spmv.cu:62-157; here it is a hand-written CDNA4 kernel).
 //
-// Strategy: "vector CSR" — a power-of-two sub-wave of W lanes per row,
-// W picked by the host from the mean row length.  Within a row the W lanes
-// read vals/indices coalesced; sub-wave reduction via __shfl_down (64-wide
-// waves, cdna_hip_programming.md §1).  fp64 CSR SpMV is memory-bound
-// (12-16 B/nnz), so the design minimizes streamed bytes:
+// Strategy: "vector CSR" — a power-of-two sub-wave of W lanes per row.
+// fp64 CSR SpMV streams 12-16 B/nnz and issues 3 loads per element
+// (val, idx, x-gather), so it can be VMEM-ISSUE-bound before it is
+// bandwidth-bound.  Design choices, each measured
+// (profiles/spmv_sweep_r01.txt and successors):
 //   - int32 column indices when the matrix allows (DISPATCH_IDX_T)
-//   - plain (cached) loads: a measured nt-load variant was 25-40%% SLOWER
-//     (profiles/spmv_sweep_r01.txt) — nt defeats the L2/L3 help on the
-//     vals/indices stream here, unlike the decode-weight case
-//   - grid-stride with a capped grid (G11), no LDS staging.
+//   - PAIR variant: each lane owns 2 consecutive elements loaded as one
+//     16-B val vector + one 8-B idx vector from an even-aligned base —
+//     4 VMEM instructions per 2 elements instead of 6
+//   - plain (cached) loads: nt measured 25-40% slower here
+//   - W = largest pow2 <= mean/2 (2 elements per lane)
+//   - optional XCD-aware bijective block remap: contiguous row chunks
+//     per XCD L2 (x reuse between neighboring rows)
 
 #include "common.h"
 
@@ -30,7 +33,15 @@ __device__ inline Cplx<T> nt_load(const Cplx<T>* p) {
           __builtin_nontemporal_load(&p->im)};
 }
 
-template <typename T, typename I, int W, bool NT>
+__device__ inline int xcd_swizzle(int b, int nwg) {
+  // bijective remap: XCD (= b % 8 by observed dispatch) gets a contiguous
+  // chunk of the grid (cdna_hip_programming.md T1, bijective variant)
+  const int q = nwg >> 3, r = nwg & 7;
+  const int xcd = b & 7, pos = b >> 3;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+}
+
+template <typename T, typename I, int W, bool NT, bool SWZ>
 __global__ __launch_bounds__(LS_THREADS) void spmv_vector_kernel(
     const i64* __restrict__ indptr, const I* __restrict__ indices,
     const T* __restrict__ vals, const T* __restrict__ x, T* __restrict__ y,
@@ -38,8 +49,9 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_vector_kernel(
   constexpr int ROWS_PER_BLOCK = LS_THREADS / W;
   const int group = threadIdx.x / W;
   const int lane = threadIdx.x % W;
+  const int blk = SWZ ? xcd_swizzle(blockIdx.x, gridDim.x) : blockIdx.x;
   const i64 stride = (i64)gridDim.x * ROWS_PER_BLOCK;
-  for (i64 row = (i64)blockIdx.x * ROWS_PER_BLOCK + group; row < n_rows;
+  for (i64 row = (i64)blk * ROWS_PER_BLOCK + group; row < n_rows;
        row += stride) {
     const i64 s = indptr[row];
     const i64 e = indptr[row + 1];
@@ -60,29 +72,89 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_vector_kernel(
   }
 }
 
+// PAIR variant (real dtypes): 2 consecutive elements per lane per step,
+// loaded as vectors from an even (16-B for f64) aligned base — halves the
+// VMEM instruction count on the vals/indices streams.
+template <typename T, typename I, int W, bool SWZ>
+__global__ __launch_bounds__(LS_THREADS) void spmv_pair_kernel(
+    const i64* __restrict__ indptr, const I* __restrict__ indices,
+    const T* __restrict__ vals, const T* __restrict__ x, T* __restrict__ y,
+    i64 n_rows, int accumulate) {
+  using T2 = __attribute__((ext_vector_type(2))) T;
+  using I2 = __attribute__((ext_vector_type(2))) I;
+  constexpr int ROWS_PER_BLOCK = LS_THREADS / W;
+  const int group = threadIdx.x / W;
+  const int lane = threadIdx.x % W;
+  const int blk = SWZ ? xcd_swizzle(blockIdx.x, gridDim.x) : blockIdx.x;
+  const i64 stride = (i64)gridDim.x * ROWS_PER_BLOCK;
+  for (i64 row = (i64)blk * ROWS_PER_BLOCK + group; row < n_rows;
+       row += stride) {
+    const i64 s = indptr[row];
+    const i64 e = indptr[row + 1];
+    const i64 s2 = s & ~(i64)1;  // aligned pair base (never < 0)
+    T acc = ls_zero<T>();
+    for (i64 pp = s2 + 2 * lane; pp < e; pp += 2 * W) {
+      const T2 v = *reinterpret_cast<const T2*>(vals + pp);
+      const I2 c = *reinterpret_cast<const I2*>(indices + pp);
+      if (pp >= s) acc += v.x * x[c.x];
+      if (pp + 1 < e) acc += v.y * x[c.y];
+    }
+    acc = group_reduce_sum<T, W>(acc);
+    if (lane == 0) {
+      if (accumulate)
+        y[row] += acc;
+      else
+        y[row] = acc;
+    }
+  }
+}
+
 template <typename T, typename I>
 void spmv_launch(const i64* indptr, const I* indices, const T* vals,
                  const T* x, T* y, i64 n_rows, i64 nnz, bool accumulate,
-                 int w_override, bool nt, hipStream_t stream) {
+                 int w_override, bool nt, int pair_mode, int swz_mode,
+                 hipStream_t stream) {
   const double mean = n_rows > 0 ? (double)nnz / (double)n_rows : 0.0;
-  // W = largest power of two <= mean/2 (measured on MI355X: ~2 elements
-  // per lane beats 1; profiles/spmv_sweep_r01.txt — W=2 gives 504 GF/s on
-  // 5-pt Poisson fp64 vs 442 at W=4).
+  // W = largest power of two <= mean/2 (measured: ~2 elements per lane,
+  // profiles/spmv_sweep_r01.txt)
   int W = 1;
   while (W < 64 && (double)(W * 4) <= mean) W *= 2;
   if (w_override > 0) W = w_override;
+  constexpr bool is_real = !is_cplx<T>::value;
+  const bool pair = (pair_mode < 0) ? (is_real && mean >= 3.0)
+                                    : (pair_mode == 1 && is_real);
+  const bool swz = swz_mode == 1;
   auto launch = [&](auto wtag) {
     constexpr int WS = decltype(wtag)::value;
     constexpr int RPB = LS_THREADS / WS;
     int grid = grid_1d(n_rows, RPB, 8192);
+    if constexpr (!is_cplx<T>::value) {
+      if (pair) {
+        if (swz)
+          hipLaunchKernelGGL((spmv_pair_kernel<T, I, WS, true>),
+                             dim3(grid), dim3(LS_THREADS), 0, stream,
+                             indptr, indices, vals, x, y, n_rows,
+                             accumulate ? 1 : 0);
+        else
+          hipLaunchKernelGGL((spmv_pair_kernel<T, I, WS, false>),
+                             dim3(grid), dim3(LS_THREADS), 0, stream,
+                             indptr, indices, vals, x, y, n_rows,
+                             accumulate ? 1 : 0);
+        return;
+      }
+    }
     if (nt)
-      hipLaunchKernelGGL((spmv_vector_kernel<T, I, WS, true>), dim3(grid),
-                         dim3(LS_THREADS), 0, stream, indptr, indices, vals,
-                         x, y, n_rows, accumulate ? 1 : 0);
+      hipLaunchKernelGGL((spmv_vector_kernel<T, I, WS, true, false>),
+                         dim3(grid), dim3(LS_THREADS), 0, stream, indptr,
+                         indices, vals, x, y, n_rows, accumulate ? 1 : 0);
+    else if (swz)
+      hipLaunchKernelGGL((spmv_vector_kernel<T, I, WS, false, true>),
+                         dim3(grid), dim3(LS_THREADS), 0, stream, indptr,
+                         indices, vals, x, y, n_rows, accumulate ? 1 : 0);
     else
-      hipLaunchKernelGGL((spmv_vector_kernel<T, I, WS, false>), dim3(grid),
-                         dim3(LS_THREADS), 0, stream, indptr, indices, vals,
-                         x, y, n_rows, accumulate ? 1 : 0);
+      hipLaunchKernelGGL((spmv_vector_kernel<T, I, WS, false, false>),
+                         dim3(grid), dim3(LS_THREADS), 0, stream, indptr,
+                         indices, vals, x, y, n_rows, accumulate ? 1 : 0);
   };
   switch (W) {
     case 1: launch(std::integral_constant<int, 1>{}); break;
@@ -101,7 +173,7 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
 void ls_spmv(uintptr_t indptr, uintptr_t indices, uintptr_t vals,
              uintptr_t x, uintptr_t y, i64 n_rows, i64 nnz, int dtype,
              int idx_dtype, bool accumulate, uintptr_t stream,
-             int w_override, bool nt) {
+             int w_override, bool nt, int pair_mode, int swz_mode) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
       spmv_launch<val_t, idx_t>(
@@ -109,5 +181,6 @@ void ls_spmv(uintptr_t indptr, uintptr_t indices, uintptr_t vals,
           reinterpret_cast<const idx_t*>(indices),
           reinterpret_cast<const val_t*>(vals),
           reinterpret_cast<const val_t*>(x), reinterpret_cast<val_t*>(y),
-          n_rows, nnz, accumulate, w_override, nt, s))));
+          n_rows, nnz, accumulate, w_override, nt, pair_mode, swz_mode,
+          s))));
 }
