@@ -55,8 +55,9 @@ def main():
     variants = []
     for w in (1, 2, 4):
         for pair in (0, 1):
-            for swz in (0, 1):
-                variants.append(("i32", A, w, pair, swz))
+            variants.append(("i32", A, w, pair, 0))
+    variants.append(("i32", A, 2, 2, 0))   # stream PE=8
+    variants.append(("i32", A, 2, 3, 0))   # stream PE=16
 
     results = {v[:1] + v[2:]: [] for v in variants}
     for r in range(args.rounds):

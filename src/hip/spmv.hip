@@ -109,6 +109,67 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_pair_kernel(
   }
 }
 
+// STREAM variant (CSR-stream): each block owns a fixed ELEMENT range
+// [base, base+NB), loads val/idx pairs coalesced, gathers x, stages the
+// products in LDS, then reduces its rows from LDS (binary search for the
+// row range).  Rows fully inside the block store directly; block-boundary
+// rows atomicAdd into a pre-zeroed y.  Short-row matrices only (long rows
+// would serialize one thread's reduce) — selection in spmv_launch.
+__device__ inline i64 last_row_leq(const i64* __restrict__ indptr,
+                                   i64 n_rows, i64 v) {
+  // max r in [0, n_rows) with indptr[r] <= v
+  i64 lo = 0, hi = n_rows - 1;
+  while (lo < hi) {
+    const i64 mid = (lo + hi + 1) >> 1;
+    if (indptr[mid] <= v)
+      lo = mid;
+    else
+      hi = mid - 1;
+  }
+  return lo;
+}
+
+template <typename T, typename I, int PE>
+__global__ __launch_bounds__(LS_THREADS) void spmv_stream_kernel(
+    const i64* __restrict__ indptr, const I* __restrict__ indices,
+    const T* __restrict__ vals, const T* __restrict__ x, T* __restrict__ y,
+    i64 n_rows, i64 nnz) {
+  using T2 = __attribute__((ext_vector_type(2))) T;
+  using I2 = __attribute__((ext_vector_type(2))) I;
+  constexpr int NB = LS_THREADS * PE;
+  __shared__ T prods[NB];
+  const i64 base = (i64)blockIdx.x * NB;
+  if (base >= nnz) return;
+  const i64 end = (base + NB < nnz) ? base + NB : nnz;
+  // load + multiply, pairs (base is NB-aligned, NB even => 16-B aligned)
+  for (int t = threadIdx.x * 2; t < NB; t += LS_THREADS * 2) {
+    const i64 j = base + t;
+    if (j + 1 < end) {
+      const T2 v = *reinterpret_cast<const T2*>(vals + j);
+      const I2 c = *reinterpret_cast<const I2*>(indices + j);
+      const T2 p = {v.x * x[c.x], v.y * x[c.y]};
+      *reinterpret_cast<T2*>(&prods[t]) = p;
+    } else if (j < end) {
+      prods[t] = vals[j] * x[indices[j]];
+    }
+  }
+  __syncthreads();
+  const i64 r0 = last_row_leq(indptr, n_rows, base);
+  const i64 r1 = last_row_leq(indptr, n_rows, end - 1);
+  for (i64 r = r0 + threadIdx.x; r <= r1; r += LS_THREADS) {
+    const i64 rs = indptr[r], re = indptr[r + 1];
+    const i64 s = rs > base ? rs : base;
+    const i64 e = re < end ? re : end;
+    T acc = ls_zero<T>();
+    for (i64 j = s; j < e; ++j) acc += prods[j - base];
+    if (rs >= base && re <= end) {
+      y[r] = acc;  // interior row: exactly one owner
+    } else if (e > s) {
+      atomicAdd(&y[r], acc);  // boundary row: y pre-zeroed by host
+    }
+  }
+}
+
 template <typename T, typename I>
 void spmv_launch(const i64* indptr, const I* indices, const T* vals,
                  const T* x, T* y, i64 n_rows, i64 nnz, bool accumulate,
@@ -124,6 +185,29 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
   const bool pair = (pair_mode < 0) ? (is_real && mean >= 3.0)
                                     : (pair_mode == 1 && is_real);
   const bool swz = swz_mode == 1;
+  // STREAM variant: pair_mode 2 (PE=8) / 3 (PE=16); real dtypes,
+  // non-accumulating only (boundary rows atomicAdd into zeroed y)
+  if constexpr (!is_cplx<T>::value) {
+    if ((pair_mode == 2 || pair_mode == 3) && !accumulate && n_rows > 0) {
+      ls_check(hipMemsetAsync(y, 0, n_rows * sizeof(T), stream),
+               "spmv memset");
+      if (pair_mode == 2) {
+        constexpr int NB = LS_THREADS * 8;
+        const i64 grid = (nnz + NB - 1) / NB;
+        hipLaunchKernelGGL((spmv_stream_kernel<T, I, 8>), dim3(grid),
+                           dim3(LS_THREADS), 0, stream, indptr, indices,
+                           vals, x, y, n_rows, nnz);
+      } else {
+        constexpr int NB = LS_THREADS * 16;
+        const i64 grid = (nnz + NB - 1) / NB;
+        hipLaunchKernelGGL((spmv_stream_kernel<T, I, 16>), dim3(grid),
+                           dim3(LS_THREADS), 0, stream, indptr, indices,
+                           vals, x, y, n_rows, nnz);
+      }
+      ls_check(hipGetLastError(), "spmv_stream");
+      return;
+    }
+  }
   auto launch = [&](auto wtag) {
     constexpr int WS = decltype(wtag)::value;
     constexpr int RPB = LS_THREADS / WS;
