@@ -294,18 +294,64 @@ class csr_array(CompressedBase):
     # ------------------------------------------------------------------
     # Core ops
     # ------------------------------------------------------------------
-    def _gather_x(self, x: torch.Tensor) -> torch.Tensor:
-        """Produce the full-length x needed by local SpMV (the explicit
-        replacement of the reference's MIN_MAX column image, csr.py:591)."""
+    def _halo_plan(self):
+        """MIN_MAX column-window halo plan (the explicit replacement of
+        the reference's image(crd, x, MIN_MAX) constraint, csr.py:591).
+
+        Each rank needs only x[min_col : max_col+1] of its local matrix;
+        the plan (cached — structure is immutable) records that window and
+        the per-rank slice exchange realizing it via all-to-allv over
+        xGMI, instead of a full x all-gather."""
+        if getattr(self, "_halo_cache", None) is not None:
+            return self._halo_cache
+        N = self._shape[1]
+        part = runtime.partition(N)
+        ws = runtime.world_size
+        if self._indices.numel():
+            mn = int(self._indices.min())
+            mx = int(self._indices.max()) + 1
+        else:
+            mn = part.lo(runtime.rank)
+            mx = mn
+        wins = comm.allgather_scalars(mn).tolist(), \
+            comm.allgather_scalars(mx).tolist()
+        mns, mxs = wins
+        lo, hi = part.lo(runtime.rank), part.hi(runtime.rank)
+        send = []
+        for r in range(ws):
+            a = max(lo, int(mns[r]))
+            b = min(hi, int(mxs[r]))
+            send.append((a - lo, max(0, b - a)))
+        recv_counts = []
+        for r in range(ws):
+            a = max(part.lo(r), mn)
+            b = min(part.hi(r), mx)
+            recv_counts.append(max(0, b - a))
+        self._halo_cache = {
+            "min_col": mn, "win": mx - mn, "send": send,
+            "recv_counts": recv_counts,
+        }
+        return self._halo_cache
+
+    def _gather_x(self, x: torch.Tensor):
+        """Produce the x window needed by local SpMV; returns
+        (x_window, col_offset) where the kernel reads x_window[c - off]."""
         N = self._shape[1]
         if x.numel() == N:
-            return x  # replicated input: nothing to move
+            return x, 0  # replicated input: nothing to move
         part = runtime.partition(N)
         if x.numel() != part.count(runtime.rank):
             raise ValueError(
                 f"x length {x.numel()} is neither global ({N}) nor the "
                 f"local shard ({part.count(runtime.rank)})")
-        return comm.allgatherv(x, part.counts())
+        if runtime.world_size == 1:
+            return x, 0
+        plan = self._halo_plan()
+        chunks = [x[s:s + c] for (s, c) in plan["send"]]
+        recv = comm.alltoallv(chunks)
+        x_win = torch.cat(recv) if len(recv) > 1 else recv[0]
+        assert x_win.numel() == plan["win"], (x_win.numel(), plan)
+        return x_win, plan["min_col"]
 
     def _coerce_vec(self, other) -> torch.Tensor:
         t = other if isinstance(other, torch.Tensor) else as_torch_1d(other)
@@ -337,7 +383,7 @@ class csr_array(CompressedBase):
         x = self._coerce_vec(t).to(dt)
         A = self if self._data.dtype == dt else self.astype(
             to_numpy_dtype(dt))
-        x_full = A._gather_x(x)
+        x_win, col_off = A._gather_x(x)
         lr = self._row_hi - self._row_lo
         if out is not None:
             if out.shape[0] != lr or out.dtype != dt:
@@ -345,7 +391,8 @@ class csr_array(CompressedBase):
             y = out.reshape(-1)
         else:
             y = None
-        y = ops.spmv(A._indptr, A._indices, A._data, x_full, y)
+        y = ops.spmv(A._indptr, A._indices, A._data, x_win, y,
+                     col_offset=col_off)
         return y.reshape(-1, 1) if squeezed else y
 
     def __matmul__(self, other):
@@ -486,25 +533,59 @@ csr_matrix = csr_array
 # ---------------------------------------------------------------------------
 def spmv(A: csr_array, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
     """y = A @ x into a preallocated local y (reference csr.py:562-593)."""
-    x_full = A._gather_x(A._coerce_vec(x))
-    return ops.spmv(A._indptr, A._indices, A._data, x_full, y.reshape(-1))
+    x_win, col_off = A._gather_x(A._coerce_vec(x))
+    return ops.spmv(A._indptr, A._indices, A._data, x_win, y.reshape(-1),
+                    col_offset=col_off)
 
 
-def _gather_full_B(B: csr_array):
-    """All-gather B's rows on every rank (v1 of the reference's SpGEMM
-    B-row image, csr.py:656-666; window optimization comes later)."""
+def _gather_B_window(A: csr_array, B: csr_array):
+    """Gather only the B-row window [min_col(A), max_col(A)+1) each rank
+    actually needs — the explicit analogue of the reference's
+    FAST_IMAGE_RANGE + MIN_MAX B-row images (csr.py:606-617, 656-666).
+    Returns (Bw_indptr, Bw_indices, Bw_vals, row_offset): kernels index
+    the windowed indptr through an offset base pointer."""
     if runtime.world_size == 1:
-        return B._indptr, B._indices, B._data
-    counts_rows = B._part.counts()
-    row_nnz = B._indptr[1:] - B._indptr[:-1]
-    all_row_nnz = comm.allgatherv(row_nnz, counts_rows)
-    g_indptr = torch.zeros(B._shape[0] + 1, dtype=_INDEX_DTYPE,
-                           device=B._data.device)
-    torch.cumsum(all_row_nnz, dim=0, out=g_indptr[1:])
-    nnz_counts = comm.allgather_scalars(B._data.numel()).tolist()
-    g_indices = comm.allgatherv(B._indices, nnz_counts)
-    g_data = comm.allgatherv(B._data, nnz_counts)
-    return g_indptr, g_indices, g_data
+        return B._indptr, B._indices, B._data, 0
+    dev = B._data.device
+    ws = runtime.world_size
+    if A._indices.numel():
+        mn = int(A._indices.min())
+        mx = int(A._indices.max()) + 1
+    else:
+        mn = mx = 0
+    mns = comm.allgather_scalars(mn).tolist()
+    mxs = comm.allgather_scalars(mx).tolist()
+    lo, hi = B._part.lo(runtime.rank), B._part.hi(runtime.rank)
+    # element offsets for all send ranges in one host transfer
+    bounds = []
+    for r in range(ws):
+        a = max(lo, int(mns[r]))
+        b = min(hi, int(mxs[r]))
+        if b < a:
+            a = b = lo
+        bounds.append((a, b))
+    probe = torch.tensor([v - lo for ab in bounds for v in ab],
+                         dtype=torch.int64, device=dev)
+    el = B._indptr[probe].cpu().tolist()
+    row_nnz_local = B._indptr[1:] - B._indptr[:-1]
+    send_rows, send_idx, send_val = [], [], []
+    for r, (a, b) in enumerate(bounds):
+        s_el, e_el = el[2 * r], el[2 * r + 1]
+        send_rows.append(row_nnz_local[a - lo:b - lo])
+        send_idx.append(B._indices[s_el:e_el])
+        send_val.append(B._data[s_el:e_el])
+    recv_rows = comm.alltoallv(send_rows)
+    recv_idx = comm.alltoallv(send_idx)
+    recv_val = comm.alltoallv(send_val)
+    row_nnz_win = torch.cat(recv_rows) if ws > 1 else recv_rows[0]
+    n_win = mx - mn
+    assert row_nnz_win.numel() == n_win, (row_nnz_win.numel(), mn, mx)
+    Bw_indptr = torch.zeros(n_win + 1, dtype=torch.int64, device=dev)
+    if n_win:
+        torch.cumsum(row_nnz_win, dim=0, out=Bw_indptr[1:])
+    Bw_indices = torch.cat(recv_idx)
+    Bw_vals = torch.cat(recv_val)
+    return Bw_indptr, Bw_indices, Bw_vals, mn
 
 
 def spgemm_csr_csr_csr(A: csr_array, B: csr_array) -> csr_array:
@@ -523,14 +604,14 @@ def spgemm_csr_csr_csr(A: csr_array, B: csr_array) -> csr_array:
         A = A.astype(to_numpy_dtype(dt))
     if B._data.dtype != dt:
         B = B.astype(to_numpy_dtype(dt))
-    B_indptr, B_indices, B_data = _gather_full_B(B)
+    B_indptr, B_indices, B_data, b_off = _gather_B_window(A, B)
     A_indices = A._indices
     if A_indices.dtype != B_indices.dtype:  # unify for the kernels
         A_indices = A_indices.to(torch.int64)
         B_indices = B_indices.to(torch.int64)
     C_indptr, C_indices, C_vals = ops.spgemm_local(
         A._indptr, A_indices, A._data, B_indptr, B_indices, B_data,
-        B.shape[1])
+        B.shape[1], b_row_offset=b_off)
     out = csr_array.__new__(csr_array)
     out._init_local(C_indptr, C_indices, C_vals, (A.shape[0], B.shape[1]))
     return out

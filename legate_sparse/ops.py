@@ -51,26 +51,31 @@ def _stream() -> int:
 def spmv(indptr: torch.Tensor, indices: torch.Tensor, vals: torch.Tensor,
          x: torch.Tensor, y: Optional[torch.Tensor] = None,
          accumulate: bool = False, w_override: int = 0,
-         nt: bool = False, pair: int = -1, swz: int = 0) -> torch.Tensor:
+         nt: bool = False, pair: int = -1, swz: int = 0,
+         col_offset: int = 0) -> torch.Tensor:
+    """col_offset: kernels read x[c - col_offset] for global column c —
+    realized as an adjusted base pointer, the same "fake offset dense
+    pointer" trick the reference plays on cuSPARSE (spmv.cu:75-90)."""
     n_rows = indptr.numel() - 1
     if y is None:
         y = torch.empty(n_rows, dtype=vals.dtype, device=vals.device)
         accumulate = False
+    x_ptr = x.data_ptr() - int(col_offset) * x.element_size()
     if _use_hip(vals):
         ext = _cext.require_hip()
         ext.spmv(indptr.data_ptr(), indices.data_ptr(), vals.data_ptr(),
-                 x.data_ptr(), y.data_ptr(), n_rows, vals.numel(),
+                 x_ptr, y.data_ptr(), n_rows, vals.numel(),
                  _code(vals), _icode(indices), accumulate, _stream(),
                  int(w_override), bool(nt), int(pair), int(swz))
         return y
     if not vals.is_cuda and _cext.has_cpu():
         _cext.require_cpu().spmv(indptr.data_ptr(), indices.data_ptr(),
-                                 vals.data_ptr(), x.data_ptr(), y.data_ptr(),
+                                 vals.data_ptr(), x_ptr, y.data_ptr(),
                                  n_rows, _code(vals), _icode(indices),
                                  accumulate)
         return y
     # torch fallback (debug / extension-less CPU)
-    prod = vals * x[indices.long()]
+    prod = vals * x[(indices.long() - col_offset)]
     row_ids = torch.repeat_interleave(
         torch.arange(n_rows, device=vals.device),
         (indptr[1:] - indptr[:-1]),
@@ -87,19 +92,23 @@ def spmv(indptr: torch.Tensor, indices: torch.Tensor, vals: torch.Tensor,
 def spgemm_local(
     A_indptr: torch.Tensor, A_indices: torch.Tensor, A_vals: torch.Tensor,
     B_indptr: torch.Tensor, B_indices: torch.Tensor, B_vals: torch.Tensor,
-    n_colsB: int,
+    n_colsB: int, b_row_offset: int = 0,
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
-    """Gustavson 2-phase; B rows indexable by A's (global) column ids."""
+    """Gustavson 2-phase.  B rows are indexed by A's GLOBAL column ids;
+    when only the window [b_row_offset, ...) of B's rows was gathered,
+    the kernels see an offset indptr base pointer (windowed B-row image,
+    reference csr.py:656-666)."""
     n_rows = A_indptr.numel() - 1
     dev = A_vals.device
+    bp_ptr = B_indptr.data_ptr() - int(b_row_offset) * 8
     if _use_hip(A_vals):
         return _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices,
-                           B_vals, n_colsB)
+                           B_vals, n_colsB, bp_ptr)
     if not A_vals.is_cuda and _cext.has_cpu():
         ext = _cext.require_cpu()
         row_nnz = torch.empty(n_rows, dtype=torch.int64)
         ext.spgemm_symbolic(A_indptr.data_ptr(), A_indices.data_ptr(),
-                            n_rows, B_indptr.data_ptr(),
+                            n_rows, bp_ptr,
                             B_indices.data_ptr(), int(n_colsB),
                             row_nnz.data_ptr(), _icode(A_indices))
         C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64)
@@ -108,7 +117,7 @@ def spgemm_local(
         C_indices = torch.empty(nnz, dtype=A_indices.dtype)
         C_vals = torch.empty(nnz, dtype=A_vals.dtype)
         ext.spgemm_numeric(A_indptr.data_ptr(), A_indices.data_ptr(),
-                           A_vals.data_ptr(), n_rows, B_indptr.data_ptr(),
+                           A_vals.data_ptr(), n_rows, bp_ptr,
                            B_indices.data_ptr(), B_vals.data_ptr(),
                            int(n_colsB), C_indptr.data_ptr(),
                            C_indices.data_ptr(), C_vals.data_ptr(),
@@ -116,11 +125,11 @@ def spgemm_local(
         return C_indptr, C_indices, C_vals
     # torch fallback: ESC (expand - sort - compress), works on any device.
     return _spgemm_esc(A_indptr, A_indices, A_vals, B_indptr, B_indices,
-                       B_vals, n_colsB)
+                       B_vals, n_colsB, b_row_offset)
 
 
 def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
-                n_colsB):
+                n_colsB, bp_ptr=None):
     """Binned hash-table Gustavson on gfx950 (src/hip/spgemm.hip).
 
     Rows are binned by their expansion upper bound so the per-row hash
@@ -137,10 +146,12 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     assert A_indices.dtype == B_indices.dtype
     if n_colsB >= 2 ** 32:
         raise NotImplementedError("SpGEMM requires B.shape[1] < 2^32")
+    if bp_ptr is None:
+        bp_ptr = B_indptr.data_ptr()
 
     ub = torch.empty(n_rows, dtype=torch.int64, device=dev)
     ext.spgemm_row_ub(A_indptr.data_ptr(), A_indices.data_ptr(),
-                      B_indptr.data_ptr(), ub.data_ptr(), n_rows, icode, st)
+                      bp_ptr, ub.data_ptr(), n_rows, icode, st)
     cuts = ext.spgemm_lds_bins  # (48, 128, 1024, 4096)
     bins = []
     prev = -1
@@ -154,7 +165,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         if rows.numel():
             ext.spgemm_symbolic_lds(
                 cfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
-                A_indices.data_ptr(), B_indptr.data_ptr(),
+                A_indices.data_ptr(), bp_ptr,
                 B_indices.data_ptr(), row_nnz.data_ptr(), icode, st)
     g_keys = g_off = g_sizes = None
     if rows_g.numel():
@@ -168,7 +179,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         g_keys = torch.full((total,), -1, dtype=torch.int32, device=dev)
         ext.spgemm_symbolic_global(
             rows_g.data_ptr(), rows_g.numel(), A_indptr.data_ptr(),
-            A_indices.data_ptr(), B_indptr.data_ptr(),
+            A_indices.data_ptr(), bp_ptr,
             B_indices.data_ptr(), g_keys.data_ptr(), g_off.data_ptr(),
             g_sizes.data_ptr(), row_nnz.data_ptr(), icode, st)
 
@@ -183,7 +194,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
             ext.spgemm_numeric_lds(
                 cfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
                 A_indices.data_ptr(), A_vals.data_ptr(),
-                B_indptr.data_ptr(), B_indices.data_ptr(),
+                bp_ptr, B_indices.data_ptr(),
                 B_vals.data_ptr(), C_indptr.data_ptr(),
                 C_indices.data_ptr(), C_vals.data_ptr(), code, icode, st)
     if rows_g.numel():
@@ -191,7 +202,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         g_vals = torch.zeros(g_keys.numel(), dtype=A_vals.dtype, device=dev)
         ext.spgemm_numeric_global(
             rows_g.data_ptr(), rows_g.numel(), A_indptr.data_ptr(),
-            A_indices.data_ptr(), A_vals.data_ptr(), B_indptr.data_ptr(),
+            A_indices.data_ptr(), A_vals.data_ptr(), bp_ptr,
             B_indices.data_ptr(), B_vals.data_ptr(), g_keys.data_ptr(),
             g_vals.data_ptr(), g_off.data_ptr(), g_sizes.data_ptr(),
             C_indptr.data_ptr(), C_indices.data_ptr(), C_vals.data_ptr(),
@@ -215,12 +226,12 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
 
 
 def _spgemm_esc(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
-                n_colsB):
+                n_colsB, b_row_offset: int = 0):
     dev = A_vals.device
     n_rows = A_indptr.numel() - 1
     a_rows = torch.repeat_interleave(
         torch.arange(n_rows, device=dev), A_indptr[1:] - A_indptr[:-1])
-    k = A_indices
+    k = A_indices.long() - b_row_offset
     blen = (B_indptr[1:] - B_indptr[:-1])[k.long()]
     e_rows = torch.repeat_interleave(a_rows, blen)
     e_avals = torch.repeat_interleave(A_vals, blen)
