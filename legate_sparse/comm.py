@@ -134,6 +134,44 @@ def alltoallv(send_chunks: List[torch.Tensor]) -> List[torch.Tensor]:
     return out
 
 
+def alltoallv_single_async(send_buf: torch.Tensor, in_splits, out_splits):
+    """Known-size all-to-all on a packed buffer; returns (recv, work).
+    nccl: async (work.wait() inserts the stream dependency) — lets the
+    caller overlap local compute with the exchange.  gloo: synchronous
+    emulation (returns work=None)."""
+    ws = _ws()
+    if ws == 1:
+        return send_buf, None
+    device = send_buf.device
+    recv = torch.empty(int(sum(out_splits)), dtype=send_buf.dtype,
+                       device=device)
+    if dist.get_backend() != "gloo":
+        work = dist.all_to_all_single(
+            recv, send_buf,
+            output_split_sizes=[int(x) for x in out_splits],
+            input_split_sizes=[int(x) for x in in_splits],
+            async_op=True)
+        return recv, work
+    # gloo: emulate with per-rank broadcast of each rank's packed buffer
+    sizes_all = [torch.zeros(ws, dtype=torch.int64) for _ in range(ws)]
+    mine = torch.tensor([int(x) for x in in_splits], dtype=torch.int64)
+    dist.all_gather(sizes_all, mine)
+    me = runtime.rank
+    off_out = 0
+    for r in range(ws):
+        buf = torch.empty(int(sizes_all[r].sum()), dtype=send_buf.dtype)
+        if r == me:
+            buf.copy_(send_buf)
+        dist.broadcast(buf, src=r)
+        offs = [0]
+        for v in sizes_all[r].tolist():
+            offs.append(offs[-1] + int(v))
+        piece = buf[offs[me]:offs[me + 1]]
+        recv[off_out:off_out + piece.numel()] = piece
+        off_out += piece.numel()
+    return recv, None
+
+
 def exclusive_scan_offsets(per_rank: torch.Tensor) -> torch.Tensor:
     """Exclusive prefix sum of an all-gathered per-rank count vector
     (the thrust::exclusive_scan of reference spgemm cu:53-54)."""

@@ -268,6 +268,7 @@ class csr_array(CompressedBase):
             raise ValueError("data size mismatch")
         _check_value_dtype(t.dtype)
         self._data = t.contiguous()
+        self._ov_cache = None  # halo/interior split caches values
 
     @property
     def indices(self) -> torch.Tensor:
@@ -353,6 +354,56 @@ class csr_array(CompressedBase):
         assert x_win.numel() == plan["win"], (x_win.numel(), plan)
         return x_win, plan["min_col"]
 
+    def _split_for_overlap(self):
+        """Cached interior/halo split of the local rows: interior columns
+        fall in this rank's own x shard (no comm), halo columns don't.
+        Enables overlapping the halo all-to-all with the interior SpMV
+        (the overlap Legion gave the reference for free — SURVEY §2.4)."""
+        if getattr(self, "_ov_cache", None) is not None:
+            return self._ov_cache
+        N = self._shape[1]
+        part = runtime.partition(N)
+        lo, hi = part.lo(runtime.rank), part.hi(runtime.rank)
+        lr = self._row_hi - self._row_lo
+        dev = self._data.device
+        idx = self._indices.long()
+        mask = (idx >= lo) & (idx < hi)
+        row_ids = torch.repeat_interleave(
+            torch.arange(lr, device=dev), self._indptr[1:] - self._indptr[:-1])
+
+        def build(m):
+            cnt = torch.bincount(row_ids[m], minlength=lr)
+            ip = torch.zeros(lr + 1, dtype=torch.int64, device=dev)
+            torch.cumsum(cnt, dim=0, out=ip[1:])
+            return ip, self._indices[m].contiguous(), \
+                self._data[m].contiguous()
+
+        self._ov_cache = (build(mask), build(~mask))
+        return self._ov_cache
+
+    def _matvec_dist(self, x: torch.Tensor,
+                     y: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Distributed matvec with comm/compute overlap: the halo
+        all-to-all runs async (RCCL stream) while the interior SpMV
+        computes on the compute stream; the halo part then accumulates."""
+        N = self._shape[1]
+        part = runtime.partition(N)
+        lo = part.lo(runtime.rank)
+        (ip_i, ix_i, dv_i), (ip_h, ix_h, dv_h) = self._split_for_overlap()
+        if ix_h.numel() == 0:
+            return ops.spmv(ip_i, ix_i, dv_i, x, y, col_offset=lo)
+        plan = self._halo_plan()
+        pieces = [x[s:s + c] for (s, c) in plan["send"] if c > 0]
+        send = torch.cat(pieces) if pieces else x[:0]
+        x_win, work = comm.alltoallv_single_async(
+            send, [c for (_, c) in plan["send"]], plan["recv_counts"])
+        y = ops.spmv(ip_i, ix_i, dv_i, x, y, col_offset=lo)
+        if work is not None:
+            work.wait()
+        ops.spmv(ip_h, ix_h, dv_h, x_win, y, accumulate=True,
+                 col_offset=plan["min_col"])
+        return y
+
     def _coerce_vec(self, other) -> torch.Tensor:
         t = other if isinstance(other, torch.Tensor) else as_torch_1d(other)
         return t.to(self._data.device).reshape(-1).contiguous()
@@ -383,7 +434,6 @@ class csr_array(CompressedBase):
         x = self._coerce_vec(t).to(dt)
         A = self if self._data.dtype == dt else self.astype(
             to_numpy_dtype(dt))
-        x_win, col_off = A._gather_x(x)
         lr = self._row_hi - self._row_lo
         if out is not None:
             if out.shape[0] != lr or out.dtype != dt:
@@ -391,8 +441,13 @@ class csr_array(CompressedBase):
             y = out.reshape(-1)
         else:
             y = None
-        y = ops.spmv(A._indptr, A._indices, A._data, x_win, y,
-                     col_offset=col_off)
+        N = self._shape[1]
+        if runtime.world_size > 1 and x.numel() != N:
+            y = A._matvec_dist(x, y)
+        else:
+            x_win, col_off = A._gather_x(x)
+            y = ops.spmv(A._indptr, A._indices, A._data, x_win, y,
+                         col_offset=col_off)
         return y.reshape(-1, 1) if squeezed else y
 
     def __matmul__(self, other):
@@ -533,7 +588,10 @@ csr_matrix = csr_array
 # ---------------------------------------------------------------------------
 def spmv(A: csr_array, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
     """y = A @ x into a preallocated local y (reference csr.py:562-593)."""
-    x_win, col_off = A._gather_x(A._coerce_vec(x))
+    xv = A._coerce_vec(x)
+    if runtime.world_size > 1 and xv.numel() != A.shape[1]:
+        return A._matvec_dist(xv, y.reshape(-1))
+    x_win, col_off = A._gather_x(xv)
     return ops.spmv(A._indptr, A._indices, A._data, x_win, y.reshape(-1),
                     col_offset=col_off)
 
