@@ -116,3 +116,67 @@ def poisson_2d(nx: int, ny: int, dtype=np.float64):
     torch.cumsum(counts, dim=0, out=indptr[1:])
     return csr_array(None, shape=(M, M),
                      _local=(indptr, indices, data))
+
+
+def rmat(scale: int, edge_factor: int = 16, seed: int = 0,
+         dtype=np.float64):
+    """Synthetic R-MAT (Graph500-style) adjacency matrix, built
+    DISTRIBUTED: each rank generates its share of edges on-device, routes
+    them to row owners via all-to-allv, and assembles local CSR rows.
+    Duplicate edges are summed (scipy COO semantics).  This is the
+    BASELINE.md SpGEMM stress operator (R-MAT scale-24)."""
+    import torch
+    from . import comm
+    from .csr import csr_array
+    from .runtime import runtime
+
+    N = 1 << scale
+    ws = runtime.world_size
+    rank = runtime.rank
+    dev = runtime.device
+    tdt = to_torch_dtype(dtype)
+    E_total = N * edge_factor
+    E = E_total // ws + (1 if rank < E_total % ws else 0)
+    gen = torch.Generator(device=dev)
+    gen.manual_seed(seed * 7919 + rank)
+    src = torch.zeros(E, dtype=torch.int64, device=dev)
+    dst = torch.zeros(E, dtype=torch.int64, device=dev)
+    # quadrant probabilities (a, b, c, d) = (0.57, 0.19, 0.19, 0.05)
+    for i in range(scale):
+        r = torch.rand(E, generator=gen, device=dev)
+        src_bit = r >= 0.76          # quadrants c + d
+        dst_bit = ((r >= 0.57) & (r < 0.76)) | (r >= 0.95)  # b + d
+        src |= src_bit.to(torch.int64) << i
+        dst |= dst_bit.to(torch.int64) << i
+    val = torch.ones(E, dtype=tdt, device=dev)
+    part = runtime.partition(N)
+    if ws > 1:
+        owner = torch.clamp(
+            torch.div(src, max(part.chunk, 1), rounding_mode="floor"),
+            max=ws - 1)
+        order = torch.argsort(owner)
+        src, dst, val = src[order], dst[order], val[order]
+        counts = torch.bincount(owner.cpu().long(), minlength=ws).tolist()
+        src = torch.cat(comm.alltoallv(list(torch.split(src, counts))))
+        dst = torch.cat(comm.alltoallv(list(torch.split(dst, counts))))
+        val = torch.cat(comm.alltoallv(list(torch.split(val, counts))))
+    lo, hi = part.lo(rank), part.hi(rank)
+    row = src - lo
+    # sort by (row, col), then sum duplicates
+    order = torch.argsort(dst, stable=True)
+    row, dst, val = row[order], dst[order], val[order]
+    order = torch.argsort(row, stable=True)
+    row, dst, val = row[order], dst[order], val[order]
+    key = row * N + dst
+    uniq, inverse = torch.unique_consecutive(key, return_inverse=True)
+    if uniq.numel() != key.numel():
+        newval = torch.zeros(uniq.numel(), dtype=tdt, device=dev)
+        newval.index_add_(0, inverse, val)
+        val = newval
+        row = torch.div(uniq, N, rounding_mode="floor")
+        dst = uniq - row * N
+    lr = hi - lo
+    cnt = torch.bincount(row, minlength=lr)
+    indptr = torch.zeros(lr + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(cnt, dim=0, out=indptr[1:])
+    return csr_array(None, shape=(N, N), _local=(indptr, dst, val))

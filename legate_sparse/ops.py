@@ -134,9 +134,14 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
 
     Rows are binned by their expansion upper bound so the per-row hash
     table fits LDS (sorted output via in-LDS bitonic); oversize rows fall
-    back to an HBM workspace table (the automatic analogue of the
-    reference's cuSPARSE ALG1/ALG3 switch, spgemm_csr_csr_csr.cu:196-216)
-    and are canonically sorted afterwards."""
+    back to an HBM workspace table and are canonically sorted afterwards.
+
+    Two modes (the analogue of the reference's cuSPARSE ALG1/ALG3 switch,
+    spgemm_csr_csr_csr.cu:196-216, selected by LS_FAST_SPGEMM):
+      exact (default): symbolic pass -> exact allocation -> numeric
+      fast:            allocate by upper bound (memory-hungry), numeric
+                       counts exact nnz, compact afterwards — skips the
+                       whole symbolic pass."""
     ext = _cext.require_hip()
     dev = A_vals.device
     st = _stream()
@@ -148,6 +153,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         raise NotImplementedError("SpGEMM requires B.shape[1] < 2^32")
     if bp_ptr is None:
         bp_ptr = B_indptr.data_ptr()
+    fast = settings.fast_spgemm
 
     ub = torch.empty(n_rows, dtype=torch.int64, device=dev)
     ext.spgemm_row_ub(A_indptr.data_ptr(), A_indices.data_ptr(),
@@ -161,12 +167,6 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     rows_g = torch.nonzero(ub > cuts[-1]).reshape(-1)
 
     row_nnz = torch.zeros(n_rows, dtype=torch.int64, device=dev)
-    for cfg, rows in enumerate(bins):
-        if rows.numel():
-            ext.spgemm_symbolic_lds(
-                cfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
-                A_indices.data_ptr(), bp_ptr,
-                B_indices.data_ptr(), row_nnz.data_ptr(), icode, st)
     g_keys = g_off = g_sizes = None
     if rows_g.numel():
         ub_g = ub[rows_g]
@@ -177,17 +177,36 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
             g_off[1:] = torch.cumsum(g_sizes[:-1], dim=0)
         total = int((g_off[-1] + g_sizes[-1]).item())
         g_keys = torch.full((total,), -1, dtype=torch.int32, device=dev)
-        ext.spgemm_symbolic_global(
-            rows_g.data_ptr(), rows_g.numel(), A_indptr.data_ptr(),
-            A_indices.data_ptr(), bp_ptr,
-            B_indices.data_ptr(), g_keys.data_ptr(), g_off.data_ptr(),
-            g_sizes.data_ptr(), row_nnz.data_ptr(), icode, st)
 
-    C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
-    torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
-    nnz = int(C_indptr[-1].item())
-    C_indices = torch.empty(nnz, dtype=A_indices.dtype, device=dev)
-    C_vals = torch.empty(nnz, dtype=A_vals.dtype, device=dev)
+    if not fast:
+        # ---- exact 2-phase: symbolic then numeric --------------------
+        for cfg, rows in enumerate(bins):
+            if rows.numel():
+                ext.spgemm_symbolic_lds(
+                    cfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
+                    A_indices.data_ptr(), bp_ptr,
+                    B_indices.data_ptr(), row_nnz.data_ptr(), icode, st)
+        if rows_g.numel():
+            ext.spgemm_symbolic_global(
+                rows_g.data_ptr(), rows_g.numel(), A_indptr.data_ptr(),
+                A_indices.data_ptr(), bp_ptr,
+                B_indices.data_ptr(), g_keys.data_ptr(), g_off.data_ptr(),
+                g_sizes.data_ptr(), row_nnz.data_ptr(), icode, st)
+        C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
+        nnz = int(C_indptr[-1].item())
+        C_indices = torch.empty(nnz, dtype=A_indices.dtype, device=dev)
+        C_vals = torch.empty(nnz, dtype=A_vals.dtype, device=dev)
+        out_indptr = C_indptr
+        nnz_ptr = 0
+    else:
+        # ---- fast: allocate by upper bound, numeric counts -----------
+        out_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(ub, dim=0, out=out_indptr[1:])
+        cap = int(out_indptr[-1].item())
+        C_indices = torch.empty(cap, dtype=A_indices.dtype, device=dev)
+        C_vals = torch.empty(cap, dtype=A_vals.dtype, device=dev)
+        nnz_ptr = row_nnz.data_ptr()
 
     for cfg, rows in enumerate(bins):
         if rows.numel():
@@ -195,8 +214,9 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 cfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
                 A_indices.data_ptr(), A_vals.data_ptr(),
                 bp_ptr, B_indices.data_ptr(),
-                B_vals.data_ptr(), C_indptr.data_ptr(),
-                C_indices.data_ptr(), C_vals.data_ptr(), code, icode, st)
+                B_vals.data_ptr(), out_indptr.data_ptr(),
+                C_indices.data_ptr(), C_vals.data_ptr(), code, icode,
+                nnz_ptr, st)
     if rows_g.numel():
         g_keys.fill_(-1)
         g_vals = torch.zeros(g_keys.numel(), dtype=A_vals.dtype, device=dev)
@@ -205,8 +225,21 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
             A_indices.data_ptr(), A_vals.data_ptr(), bp_ptr,
             B_indices.data_ptr(), B_vals.data_ptr(), g_keys.data_ptr(),
             g_vals.data_ptr(), g_off.data_ptr(), g_sizes.data_ptr(),
-            C_indptr.data_ptr(), C_indices.data_ptr(), C_vals.data_ptr(),
-            code, icode, st)
+            out_indptr.data_ptr(), C_indices.data_ptr(), C_vals.data_ptr(),
+            code, icode, nnz_ptr, st)
+
+    if fast:
+        # compact the capacity layout to exact CSR
+        C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
+        total_nnz = int(C_indptr[-1].item())
+        src = (torch.repeat_interleave(out_indptr[:-1], row_nnz)
+               + torch.arange(total_nnz, device=dev)
+               - torch.repeat_interleave(C_indptr[:-1], row_nnz))
+        C_indices = C_indices[src]
+        C_vals = C_vals[src]
+
+    if rows_g.numel():
         # canonical per-row sort for the (rare) global-bin rows only
         cnts = (C_indptr[rows_g + 1] - C_indptr[rows_g])
         total_e = int(cnts.sum().item())

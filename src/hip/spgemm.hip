@@ -132,17 +132,19 @@ __global__ __launch_bounds__(BLOCK) void spgemm_symbolic_lds_kernel(
   if (idx < n_bin_rows && lane == 0) row_nnz[rows_list[idx]] = cnt[group];
 }
 
-template <typename T, typename I, int BLOCK, int ROWS, int TBL>
+template <typename T, typename I, int BLOCK, int ROWS, int TBL,
+          bool COUNT>
 __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
     const i64* __restrict__ rows_list, i64 n_bin_rows,
     const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
     const T* __restrict__ A_vals, const i64* __restrict__ B_indptr,
     const I* __restrict__ B_indices, const T* __restrict__ B_vals,
     const i64* __restrict__ C_indptr, I* __restrict__ C_indices,
-    T* __restrict__ C_vals) {
+    T* __restrict__ C_vals, i64* __restrict__ row_nnz_out) {
   constexpr int W = BLOCK / ROWS;
   __shared__ uint32_t keys[ROWS][TBL];
   __shared__ T vals[ROWS][TBL];
+  __shared__ int cnt[ROWS];
   const int group = threadIdx.x / W;
   const int lane = threadIdx.x % W;
   const i64 idx = (i64)blockIdx.x * ROWS + group;
@@ -152,18 +154,38 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
     keys[group][t] = EMPTY;
     vals[group][t] = ls_zero<T>();
   }
+  if (COUNT && lane == 0) cnt[group] = 0;
   __syncthreads();
   if (idx < n_bin_rows) {
     const i64 row = rows_list[idx];
+    int fresh_cnt = 0;
     for (i64 jp = A_indptr[row] + lane; jp < A_indptr[row + 1]; jp += W) {
       const i64 k = (i64)A_indices[jp];
       const T a = A_vals[jp];
       for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-        const uint32_t slot = tbl_find_or_insert(
-            keys[group], TBL - 1, (uint32_t)B_indices[kp]);
-        atomic_add_val(&vals[group][slot], a * B_vals[kp]);
+        const uint32_t col = (uint32_t)B_indices[kp];
+        if constexpr (COUNT) {
+          uint32_t h = hash1(col, TBL - 1);
+          uint32_t slot;
+          while (true) {
+            uint32_t cur = keys[group][h];
+            if (cur == col) { slot = h; break; }
+            if (cur == EMPTY) {
+              uint32_t prev = atomicCAS(&keys[group][h], EMPTY, col);
+              if (prev == EMPTY) { slot = h; ++fresh_cnt; break; }
+              if (prev == col) { slot = h; break; }
+            }
+            h = (h + 1) & (TBL - 1);
+          }
+          atomic_add_val(&vals[group][slot], a * B_vals[kp]);
+        } else {
+          const uint32_t slot = tbl_find_or_insert(
+              keys[group], TBL - 1, col);
+          atomic_add_val(&vals[group][slot], a * B_vals[kp]);
+        }
       }
     }
+    if constexpr (COUNT) atomicAdd(&cnt[group], fresh_cnt);
   }
   __syncthreads();
   // in-place bitonic sort of (keys, vals) over the whole table; EMPTY
@@ -194,11 +216,12 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
   if (idx < n_bin_rows) {
     const i64 row = rows_list[idx];
     const i64 base = C_indptr[row];
-    const i64 cnt = C_indptr[row + 1] - base;
-    for (i64 t = lane; t < cnt; t += W) {
+    const i64 n_out = COUNT ? (i64)cnt[group] : C_indptr[row + 1] - base;
+    for (i64 t = lane; t < n_out; t += W) {
       C_indices[base + t] = (I)keys[group][t];
       C_vals[base + t] = vals[group][t];
     }
+    if (COUNT && lane == 0) row_nnz_out[row] = n_out;
   }
 }
 
@@ -245,7 +268,7 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_kernel(
     uint32_t* __restrict__ key_ws, T* __restrict__ val_ws,
     const i64* __restrict__ tbl_off, const i64* __restrict__ tbl_size,
     const i64* __restrict__ C_indptr, I* __restrict__ C_indices,
-    T* __restrict__ C_vals) {
+    T* __restrict__ C_vals, i64* __restrict__ row_nnz_out) {
   const i64 idx = blockIdx.x;
   if (idx >= n_bin_rows) return;
   const i64 row = rows_list[idx];
@@ -275,6 +298,8 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_kernel(
       C_vals[base + at] = vals[t];
     }
   }
+  __syncthreads();
+  if (threadIdx.x == 0 && row_nnz_out != nullptr) row_nnz_out[row] = cursor;
 }
 
 // ---------------------------------------------------------------------------
@@ -307,35 +332,39 @@ void symbolic_lds_dispatch(int cfg, const i64* rl, i64 n, const i64* ap,
   }
 }
 
-template <typename T, typename I>
+template <typename T, typename I, bool COUNT>
 void numeric_lds_dispatch(int cfg, const i64* rl, i64 n, const i64* ap,
                           const I* ai, const T* av, const i64* bp,
                           const I* bi, const T* bv, const i64* cp, I* ci,
-                          T* cv, hipStream_t s) {
+                          T* cv, i64* rn, hipStream_t s) {
   switch (cfg) {
     case 0:
-      hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, I, 256, 32, 64>),
-                         dim3((n + 31) / 32), dim3(256), 0, s, rl, n, ap,
-                         ai, av, bp, bi, bv, cp, ci, cv);
+      hipLaunchKernelGGL(
+          (spgemm_numeric_lds_kernel<T, I, 256, 32, 64, COUNT>),
+          dim3((n + 31) / 32), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi,
+          bv, cp, ci, cv, rn);
       break;
     case 1:
-      hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, I, 256, 16, 256>),
-                         dim3((n + 15) / 16), dim3(256), 0, s, rl, n, ap,
-                         ai, av, bp, bi, bv, cp, ci, cv);
+      hipLaunchKernelGGL(
+          (spgemm_numeric_lds_kernel<T, I, 256, 16, 256, COUNT>),
+          dim3((n + 15) / 16), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi,
+          bv, cp, ci, cv, rn);
       break;
     case 2:
-      hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, I, 128, 2, 2048>),
-                         dim3((n + 1) / 2), dim3(128), 0, s, rl, n, ap, ai,
-                         av, bp, bi, bv, cp, ci, cv);
+      hipLaunchKernelGGL(
+          (spgemm_numeric_lds_kernel<T, I, 128, 2, 2048, COUNT>),
+          dim3((n + 1) / 2), dim3(128), 0, s, rl, n, ap, ai, av, bp, bi,
+          bv, cp, ci, cv, rn);
       break;
     default: {
       // 16-byte values (c128) at TBL=8192 exceed 160 KiB LDS; use 4096
       // slots there (bin cutoff is ub <= 4096: the table can be exactly
       // full — linear probing still terminates).
       constexpr int TBL3 = (sizeof(T) > 8) ? 4096 : 8192;
-      hipLaunchKernelGGL((spgemm_numeric_lds_kernel<T, I, 256, 1, TBL3>),
-                         dim3(n), dim3(256), 0, s, rl, n, ap, ai, av, bp,
-                         bi, bv, cp, ci, cv);
+      hipLaunchKernelGGL(
+          (spgemm_numeric_lds_kernel<T, I, 256, 1, TBL3, COUNT>),
+          dim3(n), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi, bv, cp, ci,
+          cv, rn);
     }
   }
 }
@@ -381,20 +410,36 @@ void ls_spgemm_numeric_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
                            uintptr_t B_indices, uintptr_t B_vals,
                            uintptr_t C_indptr, uintptr_t C_indices,
                            uintptr_t C_vals, int dtype, int idx_dtype,
-                           uintptr_t stream) {
+                           uintptr_t row_nnz_out, uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
-  DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
-      numeric_lds_dispatch<val_t, idx_t>(
-          cfg, reinterpret_cast<const i64*>(rows_list), n_bin_rows,
-          reinterpret_cast<const i64*>(A_indptr),
-          reinterpret_cast<const idx_t*>(A_indices),
-          reinterpret_cast<const val_t*>(A_vals),
-          reinterpret_cast<const i64*>(B_indptr),
-          reinterpret_cast<const idx_t*>(B_indices),
-          reinterpret_cast<const val_t*>(B_vals),
-          reinterpret_cast<const i64*>(C_indptr),
-          reinterpret_cast<idx_t*>(C_indices),
-          reinterpret_cast<val_t*>(C_vals), s))));
+  if (row_nnz_out) {
+    DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
+        numeric_lds_dispatch<val_t, idx_t, true>(
+            cfg, reinterpret_cast<const i64*>(rows_list), n_bin_rows,
+            reinterpret_cast<const i64*>(A_indptr),
+            reinterpret_cast<const idx_t*>(A_indices),
+            reinterpret_cast<const val_t*>(A_vals),
+            reinterpret_cast<const i64*>(B_indptr),
+            reinterpret_cast<const idx_t*>(B_indices),
+            reinterpret_cast<const val_t*>(B_vals),
+            reinterpret_cast<const i64*>(C_indptr),
+            reinterpret_cast<idx_t*>(C_indices),
+            reinterpret_cast<val_t*>(C_vals),
+            reinterpret_cast<i64*>(row_nnz_out), s))));
+  } else {
+    DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
+        numeric_lds_dispatch<val_t, idx_t, false>(
+            cfg, reinterpret_cast<const i64*>(rows_list), n_bin_rows,
+            reinterpret_cast<const i64*>(A_indptr),
+            reinterpret_cast<const idx_t*>(A_indices),
+            reinterpret_cast<const val_t*>(A_vals),
+            reinterpret_cast<const i64*>(B_indptr),
+            reinterpret_cast<const idx_t*>(B_indices),
+            reinterpret_cast<const val_t*>(B_vals),
+            reinterpret_cast<const i64*>(C_indptr),
+            reinterpret_cast<idx_t*>(C_indices),
+            reinterpret_cast<val_t*>(C_vals), nullptr, s))));
+  }
   ls_check(hipGetLastError(), "spgemm_numeric_lds");
 }
 
@@ -427,7 +472,7 @@ void ls_spgemm_numeric_global(uintptr_t rows_list, i64 n_bin_rows,
                               uintptr_t tbl_off, uintptr_t tbl_size,
                               uintptr_t C_indptr, uintptr_t C_indices,
                               uintptr_t C_vals, int dtype, int idx_dtype,
-                              uintptr_t stream) {
+                              uintptr_t row_nnz_out, uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
       (spgemm_numeric_global_kernel<val_t, idx_t>), dim3(n_bin_rows),
@@ -444,6 +489,7 @@ void ls_spgemm_numeric_global(uintptr_t rows_list, i64 n_bin_rows,
       reinterpret_cast<const i64*>(tbl_size),
       reinterpret_cast<const i64*>(C_indptr),
       reinterpret_cast<idx_t*>(C_indices),
-      reinterpret_cast<val_t*>(C_vals))));
+      reinterpret_cast<val_t*>(C_vals),
+      reinterpret_cast<i64*>(row_nnz_out))));
   ls_check(hipGetLastError(), "spgemm_numeric_global");
 }

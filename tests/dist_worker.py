@@ -108,6 +108,25 @@ def main():
     dist.all_reduce(t)
     check("diags", np.allclose(t.numpy(), want.toarray()))
 
+    # --- R-MAT distributed generation + SpGEMM ---
+    from legate_sparse.gallery import rmat
+    Ar = rmat(8, edge_factor=6, seed=5)
+    r_, c_, v_ = Ar.tocoo()
+    import scipy.sparse as _sp
+    local = _sp.coo_array((to_np(v_), (to_np(r_), to_np(c_))),
+                          shape=Ar.shape).toarray()
+    t = torch.from_numpy(local)
+    dist.all_reduce(t)
+    dense = t.numpy()
+    Cr = Ar @ Ar
+    got = np.zeros(Ar.shape)
+    lo, hi = Cr.row_range
+    got[lo:hi] = to_np(Cr.todense())
+    t2 = torch.from_numpy(got)
+    dist.all_reduce(t2)
+    check("rmat spgemm", np.allclose(t2.numpy(), dense @ dense,
+                                     rtol=1e-10, atol=1e-10))
+
     if runtime.rank == 0:
         print("DIST_ALL_OK", flush=True)
 

@@ -219,3 +219,33 @@ def test_gpu_ufuncs_and_scalars():
     want.data = np.tanh(2.0 * S.data)
     np.testing.assert_allclose(to_np(got.todense()), want.toarray(),
                                rtol=1e-12)
+
+
+def test_gpu_spgemm_fast_mode(monkeypatch):
+    """LS_FAST_SPGEMM (ALG1-style: allocate by upper bound, skip
+    symbolic) must give identical results to the exact 2-phase path."""
+    monkeypatch.setenv("LS_FAST_SPGEMM", "1")
+    n = 800
+    rng = np.random.default_rng(21)
+    rows, cols, vals = [], [], []
+    for i in range(n):
+        nnz = n if i < 3 else (60 if i < 30 else (12 if i < 300 else 3))
+        c = rng.choice(n, size=min(nnz, n), replace=False)
+        rows.extend([i] * len(c))
+        cols.extend(c.tolist())
+        vals.extend(rng.standard_normal(len(c)).tolist())
+    S = sp.csr_array((vals, (rows, cols)), shape=(n, n))
+    C = lsp.csr_array(S) @ lsp.csr_array(S)
+    np.testing.assert_allclose(to_np(C.todense()), (S @ S).toarray(),
+                               rtol=1e-10, atol=1e-10)
+
+
+def test_gpu_rmat_spgemm():
+    """R-MAT adjacency squared (power-law rows span every bin)."""
+    from legate_sparse.gallery import rmat
+    A = rmat(12, edge_factor=8, seed=3)
+    r, c, v = A.tocoo()
+    S = sp.csr_array((to_np(v), (to_np(r), to_np(c))), shape=A.shape)
+    C = A @ A
+    np.testing.assert_allclose(to_np(C.todense()), (S @ S).toarray(),
+                               rtol=1e-9, atol=1e-9)
