@@ -167,16 +167,32 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     rows_g = torch.nonzero(ub > cuts[-1]).reshape(-1)
 
     row_nnz = torch.zeros(n_rows, dtype=torch.int64, device=dev)
-    g_keys = g_off = g_sizes = None
+    g_batches = []
+    g_keys = None
     if rows_g.numel():
-        ub_g = ub[rows_g]
-        g_sizes = torch.pow(
-            2.0, torch.ceil(torch.log2(2.0 * ub_g.double()))).to(torch.int64)
-        g_off = torch.zeros_like(g_sizes)
-        if g_sizes.numel() > 1:
-            g_off[1:] = torch.cumsum(g_sizes[:-1], dim=0)
-        total = int((g_off[-1] + g_sizes[-1]).item())
-        g_keys = torch.full((total,), -1, dtype=torch.int32, device=dev)
+        # table size per row: a row has at most min(ub, n_colsB) distinct
+        # columns; clamp, then split into batches whose total workspace
+        # stays bounded (power-law matrices would otherwise demand
+        # sum-of-flops-sized tables — the R-MAT OOM)
+        need = torch.clamp(2 * ub[rows_g], max=2 * int(n_colsB)).double()
+        sizes_all = torch.pow(2.0, torch.ceil(torch.log2(need))).to(
+            torch.int64)
+        budget = max(int(sizes_all.max().item()), 1 << 27)  # >= 512 MB keys
+        csum = torch.cumsum(sizes_all, dim=0)
+        batch_id = torch.div(csum - sizes_all, budget,
+                             rounding_mode="floor")
+        n_batches = int(batch_id[-1].item()) + 1
+        max_total = 0
+        for b in range(n_batches):
+            m = batch_id == b
+            rows_b = rows_g[m]
+            sizes_b = sizes_all[m]
+            off_b = torch.cumsum(sizes_b, dim=0) - sizes_b
+            total_b = int(sizes_b.sum().item())
+            max_total = max(max_total, total_b)
+            g_batches.append((rows_b, off_b.contiguous(),
+                              sizes_b.contiguous(), total_b))
+        g_keys = torch.empty(max_total, dtype=torch.int32, device=dev)
 
     if not fast:
         # ---- exact 2-phase: symbolic then numeric --------------------
@@ -186,12 +202,13 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                     cfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
                     A_indices.data_ptr(), bp_ptr,
                     B_indices.data_ptr(), row_nnz.data_ptr(), icode, st)
-        if rows_g.numel():
+        for rows_b, off_b, sizes_b, total_b in g_batches:
+            g_keys[:total_b].fill_(-1)
             ext.spgemm_symbolic_global(
-                rows_g.data_ptr(), rows_g.numel(), A_indptr.data_ptr(),
+                rows_b.data_ptr(), rows_b.numel(), A_indptr.data_ptr(),
                 A_indices.data_ptr(), bp_ptr,
-                B_indices.data_ptr(), g_keys.data_ptr(), g_off.data_ptr(),
-                g_sizes.data_ptr(), row_nnz.data_ptr(), icode, st)
+                B_indices.data_ptr(), g_keys.data_ptr(), off_b.data_ptr(),
+                sizes_b.data_ptr(), row_nnz.data_ptr(), icode, st)
         C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
         torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
         nnz = int(C_indptr[-1].item())
@@ -201,8 +218,10 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         nnz_ptr = 0
     else:
         # ---- fast: allocate by upper bound, numeric counts -----------
+        # (clamped: a row has at most n_colsB distinct columns)
         out_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
-        torch.cumsum(ub, dim=0, out=out_indptr[1:])
+        torch.cumsum(torch.clamp(ub, max=int(n_colsB)), dim=0,
+                     out=out_indptr[1:])
         cap = int(out_indptr[-1].item())
         C_indices = torch.empty(cap, dtype=A_indices.dtype, device=dev)
         C_vals = torch.empty(cap, dtype=A_vals.dtype, device=dev)
@@ -217,16 +236,18 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 B_vals.data_ptr(), out_indptr.data_ptr(),
                 C_indices.data_ptr(), C_vals.data_ptr(), code, icode,
                 nnz_ptr, st)
-    if rows_g.numel():
-        g_keys.fill_(-1)
-        g_vals = torch.zeros(g_keys.numel(), dtype=A_vals.dtype, device=dev)
-        ext.spgemm_numeric_global(
-            rows_g.data_ptr(), rows_g.numel(), A_indptr.data_ptr(),
-            A_indices.data_ptr(), A_vals.data_ptr(), bp_ptr,
-            B_indices.data_ptr(), B_vals.data_ptr(), g_keys.data_ptr(),
-            g_vals.data_ptr(), g_off.data_ptr(), g_sizes.data_ptr(),
-            out_indptr.data_ptr(), C_indices.data_ptr(), C_vals.data_ptr(),
-            code, icode, nnz_ptr, st)
+    if g_batches:
+        g_vals = torch.empty(g_keys.numel(), dtype=A_vals.dtype, device=dev)
+        for rows_b, off_b, sizes_b, total_b in g_batches:
+            g_keys[:total_b].fill_(-1)
+            g_vals[:total_b].zero_()
+            ext.spgemm_numeric_global(
+                rows_b.data_ptr(), rows_b.numel(), A_indptr.data_ptr(),
+                A_indices.data_ptr(), A_vals.data_ptr(), bp_ptr,
+                B_indices.data_ptr(), B_vals.data_ptr(), g_keys.data_ptr(),
+                g_vals.data_ptr(), off_b.data_ptr(), sizes_b.data_ptr(),
+                out_indptr.data_ptr(), C_indices.data_ptr(),
+                C_vals.data_ptr(), code, icode, nnz_ptr, st)
 
     if fast:
         # compact the capacity layout to exact CSR
