@@ -182,6 +182,8 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         batch_id = torch.div(csum - sizes_all, budget,
                              rounding_mode="floor")
         n_batches = int(batch_id[-1].item()) + 1
+        CH = int(ext.spgemm_global_chunk)
+        a_len_g = A_indptr[rows_g + 1] - A_indptr[rows_g]
         max_total = 0
         for b in range(n_batches):
             m = batch_id == b
@@ -190,8 +192,19 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
             off_b = torch.cumsum(sizes_b, dim=0) - sizes_b
             total_b = int(sizes_b.sum().item())
             max_total = max(max_total, total_b)
+            # chunk each row's A-range so hub rows span many workgroups
+            nch = torch.div(a_len_g[m] + (CH - 1), CH,
+                            rounding_mode="floor").clamp(min=1)
+            tot_ch = int(nch.sum().item())
+            ch_rowidx = torch.repeat_interleave(
+                torch.arange(rows_b.numel(), device=dev), nch)
+            ch_start = torch.cumsum(nch, 0) - nch
+            ch_ord = (torch.arange(tot_ch, device=dev)
+                      - torch.repeat_interleave(ch_start, nch))
             g_batches.append((rows_b, off_b.contiguous(),
-                              sizes_b.contiguous(), total_b))
+                              sizes_b.contiguous(), total_b,
+                              ch_rowidx.contiguous(), ch_ord.contiguous(),
+                              tot_ch))
         g_keys = torch.empty(max_total, dtype=torch.int32, device=dev)
 
     if not fast:
@@ -202,11 +215,12 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                     cfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
                     A_indices.data_ptr(), bp_ptr,
                     B_indices.data_ptr(), row_nnz.data_ptr(), icode, st)
-        for rows_b, off_b, sizes_b, total_b in g_batches:
+        for (rows_b, off_b, sizes_b, total_b, ch_ri, ch_ord,
+             tot_ch) in g_batches:
             g_keys[:total_b].fill_(-1)
             ext.spgemm_symbolic_global(
-                rows_b.data_ptr(), rows_b.numel(), A_indptr.data_ptr(),
-                A_indices.data_ptr(), bp_ptr,
+                rows_b.data_ptr(), ch_ri.data_ptr(), ch_ord.data_ptr(),
+                tot_ch, A_indptr.data_ptr(), A_indices.data_ptr(), bp_ptr,
                 B_indices.data_ptr(), g_keys.data_ptr(), off_b.data_ptr(),
                 sizes_b.data_ptr(), row_nnz.data_ptr(), icode, st)
         C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
@@ -238,16 +252,21 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 nnz_ptr, st)
     if g_batches:
         g_vals = torch.empty(g_keys.numel(), dtype=A_vals.dtype, device=dev)
-        for rows_b, off_b, sizes_b, total_b in g_batches:
+        for (rows_b, off_b, sizes_b, total_b, ch_ri, ch_ord,
+             tot_ch) in g_batches:
             g_keys[:total_b].fill_(-1)
             g_vals[:total_b].zero_()
-            ext.spgemm_numeric_global(
-                rows_b.data_ptr(), rows_b.numel(), A_indptr.data_ptr(),
-                A_indices.data_ptr(), A_vals.data_ptr(), bp_ptr,
-                B_indices.data_ptr(), B_vals.data_ptr(), g_keys.data_ptr(),
+            ext.spgemm_numeric_global_fill(
+                rows_b.data_ptr(), ch_ri.data_ptr(), ch_ord.data_ptr(),
+                tot_ch, A_indptr.data_ptr(), A_indices.data_ptr(),
+                A_vals.data_ptr(), bp_ptr, B_indices.data_ptr(),
+                B_vals.data_ptr(), g_keys.data_ptr(), g_vals.data_ptr(),
+                off_b.data_ptr(), sizes_b.data_ptr(), code, icode, st)
+            ext.spgemm_global_compact(
+                rows_b.data_ptr(), rows_b.numel(), g_keys.data_ptr(),
                 g_vals.data_ptr(), off_b.data_ptr(), sizes_b.data_ptr(),
                 out_indptr.data_ptr(), C_indices.data_ptr(),
-                C_vals.data_ptr(), code, icode, nnz_ptr, st)
+                C_vals.data_ptr(), nnz_ptr, code, icode, st)
 
     if fast:
         # compact the capacity layout to exact CSR

@@ -231,24 +231,33 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
 // vals pre-zeroed by the host).  Output UNSORTED; host post-sorts these
 // rare rows.
 // ---------------------------------------------------------------------------
+// Hub rows are CHUNKED over many workgroups (one workgroup per
+// GLOBAL_CHUNK A-elements of a row) so a power-law row parallelizes
+// across the chip; a separate compact kernel then walks each row's table.
+constexpr int GLOBAL_CHUNK = 2048;
+
 template <typename I>
 __global__ __launch_bounds__(LS_THREADS) void spgemm_symbolic_global_kernel(
-    const i64* __restrict__ rows_list, i64 n_bin_rows,
-    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
-    const i64* __restrict__ B_indptr, const I* __restrict__ B_indices,
-    uint32_t* __restrict__ tbl_ws, const i64* __restrict__ tbl_off,
-    const i64* __restrict__ tbl_size, i64* __restrict__ row_nnz) {
-  const i64 idx = blockIdx.x;
-  if (idx >= n_bin_rows) return;
+    const i64* __restrict__ rows_list,
+    const i64* __restrict__ chunk_rowidx, const i64* __restrict__ chunk_ord,
+    i64 n_chunks, const i64* __restrict__ A_indptr,
+    const I* __restrict__ A_indices, const i64* __restrict__ B_indptr,
+    const I* __restrict__ B_indices, uint32_t* __restrict__ tbl_ws,
+    const i64* __restrict__ tbl_off, const i64* __restrict__ tbl_size,
+    i64* __restrict__ row_nnz) {
+  const i64 ch = blockIdx.x;
+  if (ch >= n_chunks) return;
+  const i64 idx = chunk_rowidx[ch];
   const i64 row = rows_list[idx];
   uint32_t* tbl = tbl_ws + tbl_off[idx];
   const uint32_t mask = (uint32_t)tbl_size[idx] - 1;
+  const i64 a0 = A_indptr[row] + chunk_ord[ch] * GLOBAL_CHUNK;
+  const i64 a1 = min(a0 + GLOBAL_CHUNK, A_indptr[row + 1]);
   __shared__ int cnt;
   if (threadIdx.x == 0) cnt = 0;
   __syncthreads();
   int local = 0;
-  for (i64 jp = A_indptr[row] + threadIdx.x; jp < A_indptr[row + 1];
-       jp += blockDim.x) {
+  for (i64 jp = a0 + threadIdx.x; jp < a1; jp += blockDim.x) {
     const i64 k = (i64)A_indices[jp];
     for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
       if (tbl_insert(tbl, mask, (uint32_t)B_indices[kp])) ++local;
@@ -256,30 +265,31 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_symbolic_global_kernel(
   }
   atomicAdd(&cnt, local);
   __syncthreads();
-  if (threadIdx.x == 0) row_nnz[row] = cnt;
+  if (threadIdx.x == 0 && cnt)
+    atomicAdd(reinterpret_cast<unsigned long long*>(&row_nnz[row]),
+              (unsigned long long)cnt);
 }
 
 template <typename T, typename I>
-__global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_kernel(
-    const i64* __restrict__ rows_list, i64 n_bin_rows,
-    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
-    const T* __restrict__ A_vals, const i64* __restrict__ B_indptr,
-    const I* __restrict__ B_indices, const T* __restrict__ B_vals,
-    uint32_t* __restrict__ key_ws, T* __restrict__ val_ws,
-    const i64* __restrict__ tbl_off, const i64* __restrict__ tbl_size,
-    const i64* __restrict__ C_indptr, I* __restrict__ C_indices,
-    T* __restrict__ C_vals, i64* __restrict__ row_nnz_out) {
-  const i64 idx = blockIdx.x;
-  if (idx >= n_bin_rows) return;
+__global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_fill(
+    const i64* __restrict__ rows_list,
+    const i64* __restrict__ chunk_rowidx, const i64* __restrict__ chunk_ord,
+    i64 n_chunks, const i64* __restrict__ A_indptr,
+    const I* __restrict__ A_indices, const T* __restrict__ A_vals,
+    const i64* __restrict__ B_indptr, const I* __restrict__ B_indices,
+    const T* __restrict__ B_vals, uint32_t* __restrict__ key_ws,
+    T* __restrict__ val_ws, const i64* __restrict__ tbl_off,
+    const i64* __restrict__ tbl_size) {
+  const i64 ch = blockIdx.x;
+  if (ch >= n_chunks) return;
+  const i64 idx = chunk_rowidx[ch];
   const i64 row = rows_list[idx];
   uint32_t* keys = key_ws + tbl_off[idx];
   T* vals = val_ws + tbl_off[idx];
   const uint32_t mask = (uint32_t)tbl_size[idx] - 1;
-  __shared__ int cursor;
-  if (threadIdx.x == 0) cursor = 0;
-  __syncthreads();
-  for (i64 jp = A_indptr[row] + threadIdx.x; jp < A_indptr[row + 1];
-       jp += blockDim.x) {
+  const i64 a0 = A_indptr[row] + chunk_ord[ch] * GLOBAL_CHUNK;
+  const i64 a1 = min(a0 + GLOBAL_CHUNK, A_indptr[row + 1]);
+  for (i64 jp = a0 + threadIdx.x; jp < a1; jp += blockDim.x) {
     const i64 k = (i64)A_indices[jp];
     const T a = A_vals[jp];
     for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
@@ -288,9 +298,26 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_kernel(
       atomic_add_val(&vals[slot], a * B_vals[kp]);
     }
   }
+}
+
+template <typename T, typename I>
+__global__ __launch_bounds__(LS_THREADS) void spgemm_global_compact(
+    const i64* __restrict__ rows_list, i64 n_bin_rows,
+    const uint32_t* __restrict__ key_ws, const T* __restrict__ val_ws,
+    const i64* __restrict__ tbl_off, const i64* __restrict__ tbl_size,
+    const i64* __restrict__ C_indptr, I* __restrict__ C_indices,
+    T* __restrict__ C_vals, i64* __restrict__ row_nnz_out) {
+  const i64 idx = blockIdx.x;
+  if (idx >= n_bin_rows) return;
+  const i64 row = rows_list[idx];
+  const uint32_t* keys = key_ws + tbl_off[idx];
+  const T* vals = val_ws + tbl_off[idx];
+  const i64 sz = tbl_size[idx];
+  __shared__ int cursor;
+  if (threadIdx.x == 0) cursor = 0;
   __syncthreads();
   const i64 base = C_indptr[row];
-  for (i64 t = threadIdx.x; t <= (i64)mask; t += blockDim.x) {
+  for (i64 t = threadIdx.x; t < sz; t += blockDim.x) {
     const uint32_t c = keys[t];
     if (c != EMPTY) {
       const int at = atomicAdd(&cursor, 1);
@@ -443,7 +470,8 @@ void ls_spgemm_numeric_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
   ls_check(hipGetLastError(), "spgemm_numeric_lds");
 }
 
-void ls_spgemm_symbolic_global(uintptr_t rows_list, i64 n_bin_rows,
+void ls_spgemm_symbolic_global(uintptr_t rows_list, uintptr_t chunk_rowidx,
+                               uintptr_t chunk_ord, i64 n_chunks,
                                uintptr_t A_indptr, uintptr_t A_indices,
                                uintptr_t B_indptr, uintptr_t B_indices,
                                uintptr_t tbl_ws, uintptr_t tbl_off,
@@ -451,9 +479,11 @@ void ls_spgemm_symbolic_global(uintptr_t rows_list, i64 n_bin_rows,
                                int idx_dtype, uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
-      (spgemm_symbolic_global_kernel<idx_t>), dim3(n_bin_rows),
+      (spgemm_symbolic_global_kernel<idx_t>), dim3(n_chunks),
       dim3(LS_THREADS), 0, s, reinterpret_cast<const i64*>(rows_list),
-      n_bin_rows, reinterpret_cast<const i64*>(A_indptr),
+      reinterpret_cast<const i64*>(chunk_rowidx),
+      reinterpret_cast<const i64*>(chunk_ord), n_chunks,
+      reinterpret_cast<const i64*>(A_indptr),
       reinterpret_cast<const idx_t*>(A_indices),
       reinterpret_cast<const i64*>(B_indptr),
       reinterpret_cast<const idx_t*>(B_indices),
@@ -464,20 +494,23 @@ void ls_spgemm_symbolic_global(uintptr_t rows_list, i64 n_bin_rows,
   ls_check(hipGetLastError(), "spgemm_symbolic_global");
 }
 
-void ls_spgemm_numeric_global(uintptr_t rows_list, i64 n_bin_rows,
-                              uintptr_t A_indptr, uintptr_t A_indices,
-                              uintptr_t A_vals, uintptr_t B_indptr,
-                              uintptr_t B_indices, uintptr_t B_vals,
-                              uintptr_t key_ws, uintptr_t val_ws,
-                              uintptr_t tbl_off, uintptr_t tbl_size,
-                              uintptr_t C_indptr, uintptr_t C_indices,
-                              uintptr_t C_vals, int dtype, int idx_dtype,
-                              uintptr_t row_nnz_out, uintptr_t stream) {
+void ls_spgemm_numeric_global_fill(uintptr_t rows_list,
+                                   uintptr_t chunk_rowidx,
+                                   uintptr_t chunk_ord, i64 n_chunks,
+                                   uintptr_t A_indptr, uintptr_t A_indices,
+                                   uintptr_t A_vals, uintptr_t B_indptr,
+                                   uintptr_t B_indices, uintptr_t B_vals,
+                                   uintptr_t key_ws, uintptr_t val_ws,
+                                   uintptr_t tbl_off, uintptr_t tbl_size,
+                                   int dtype, int idx_dtype,
+                                   uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
-      (spgemm_numeric_global_kernel<val_t, idx_t>), dim3(n_bin_rows),
+      (spgemm_numeric_global_fill<val_t, idx_t>), dim3(n_chunks),
       dim3(LS_THREADS), 0, s, reinterpret_cast<const i64*>(rows_list),
-      n_bin_rows, reinterpret_cast<const i64*>(A_indptr),
+      reinterpret_cast<const i64*>(chunk_rowidx),
+      reinterpret_cast<const i64*>(chunk_ord), n_chunks,
+      reinterpret_cast<const i64*>(A_indptr),
       reinterpret_cast<const idx_t*>(A_indices),
       reinterpret_cast<const val_t*>(A_vals),
       reinterpret_cast<const i64*>(B_indptr),
@@ -486,10 +519,27 @@ void ls_spgemm_numeric_global(uintptr_t rows_list, i64 n_bin_rows,
       reinterpret_cast<uint32_t*>(key_ws),
       reinterpret_cast<val_t*>(val_ws),
       reinterpret_cast<const i64*>(tbl_off),
+      reinterpret_cast<const i64*>(tbl_size))));
+  ls_check(hipGetLastError(), "spgemm_numeric_global_fill");
+}
+
+void ls_spgemm_global_compact(uintptr_t rows_list, i64 n_bin_rows,
+                              uintptr_t key_ws, uintptr_t val_ws,
+                              uintptr_t tbl_off, uintptr_t tbl_size,
+                              uintptr_t C_indptr, uintptr_t C_indices,
+                              uintptr_t C_vals, uintptr_t row_nnz_out,
+                              int dtype, int idx_dtype, uintptr_t stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
+      (spgemm_global_compact<val_t, idx_t>), dim3(n_bin_rows),
+      dim3(LS_THREADS), 0, s, reinterpret_cast<const i64*>(rows_list),
+      n_bin_rows, reinterpret_cast<const uint32_t*>(key_ws),
+      reinterpret_cast<const val_t*>(val_ws),
+      reinterpret_cast<const i64*>(tbl_off),
       reinterpret_cast<const i64*>(tbl_size),
       reinterpret_cast<const i64*>(C_indptr),
       reinterpret_cast<idx_t*>(C_indices),
       reinterpret_cast<val_t*>(C_vals),
       reinterpret_cast<i64*>(row_nnz_out))));
-  ls_check(hipGetLastError(), "spgemm_numeric_global");
+  ls_check(hipGetLastError(), "spgemm_global_compact");
 }
