@@ -174,37 +174,48 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         # columns; clamp, then split into batches whose total workspace
         # stays bounded (power-law matrices would otherwise demand
         # sum-of-flops-sized tables — the R-MAT OOM)
-        need = torch.clamp(2 * ub[rows_g], max=2 * int(n_colsB)).double()
+        p2n = 1
+        while p2n < int(n_colsB):
+            p2n *= 2
+        need = torch.clamp(2 * ub[rows_g], max=p2n).double()
         sizes_all = torch.pow(2.0, torch.ceil(torch.log2(need))).to(
             torch.int64)
+        # rows whose table spans all columns use IDENTITY hashing
+        # (slot = col): no probe walks, sequential-ish access
+        ident_all = sizes_all >= p2n
         budget = max(int(sizes_all.max().item()), 1 << 27)  # >= 512 MB keys
-        csum = torch.cumsum(sizes_all, dim=0)
-        batch_id = torch.div(csum - sizes_all, budget,
-                             rounding_mode="floor")
-        n_batches = int(batch_id[-1].item()) + 1
         CH = int(ext.spgemm_global_chunk)
         a_len_g = A_indptr[rows_g + 1] - A_indptr[rows_g]
         max_total = 0
-        for b in range(n_batches):
-            m = batch_id == b
-            rows_b = rows_g[m]
-            sizes_b = sizes_all[m]
-            off_b = torch.cumsum(sizes_b, dim=0) - sizes_b
-            total_b = int(sizes_b.sum().item())
-            max_total = max(max_total, total_b)
-            # chunk each row's A-range so hub rows span many workgroups
-            nch = torch.div(a_len_g[m] + (CH - 1), CH,
-                            rounding_mode="floor").clamp(min=1)
-            tot_ch = int(nch.sum().item())
-            ch_rowidx = torch.repeat_interleave(
-                torch.arange(rows_b.numel(), device=dev), nch)
-            ch_start = torch.cumsum(nch, 0) - nch
-            ch_ord = (torch.arange(tot_ch, device=dev)
-                      - torch.repeat_interleave(ch_start, nch))
-            g_batches.append((rows_b, off_b.contiguous(),
-                              sizes_b.contiguous(), total_b,
-                              ch_rowidx.contiguous(), ch_ord.contiguous(),
-                              tot_ch))
+        for ident in (False, True):
+            sel = torch.nonzero(ident_all == ident).reshape(-1)
+            if not sel.numel():
+                continue
+            sizes_s = sizes_all[sel]
+            csum = torch.cumsum(sizes_s, dim=0)
+            batch_id = torch.div(csum - sizes_s, budget,
+                                 rounding_mode="floor")
+            for b in range(int(batch_id[-1].item()) + 1):
+                m = batch_id == b
+                rows_b = rows_g[sel[m]]
+                sizes_b = sizes_s[m]
+                off_b = torch.cumsum(sizes_b, dim=0) - sizes_b
+                total_b = int(sizes_b.sum().item())
+                max_total = max(max_total, total_b)
+                # chunk rows' A-ranges so hub rows span many workgroups
+                nch = torch.div(a_len_g[sel[m]] + (CH - 1), CH,
+                                rounding_mode="floor").clamp(min=1)
+                tot_ch = int(nch.sum().item())
+                ch_rowidx = torch.repeat_interleave(
+                    torch.arange(rows_b.numel(), device=dev), nch)
+                ch_start = torch.cumsum(nch, 0) - nch
+                ch_ord = (torch.arange(tot_ch, device=dev)
+                          - torch.repeat_interleave(ch_start, nch))
+                g_batches.append((rows_b, off_b.contiguous(),
+                                  sizes_b.contiguous(), total_b,
+                                  ch_rowidx.contiguous(),
+                                  ch_ord.contiguous(), tot_ch,
+                                  1 if ident else 0))
         g_keys = torch.empty(max_total, dtype=torch.int32, device=dev)
 
     if not fast:
@@ -216,13 +227,13 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                     A_indices.data_ptr(), bp_ptr,
                     B_indices.data_ptr(), row_nnz.data_ptr(), icode, st)
         for (rows_b, off_b, sizes_b, total_b, ch_ri, ch_ord,
-             tot_ch) in g_batches:
+             tot_ch, ident) in g_batches:
             g_keys[:total_b].fill_(-1)
             ext.spgemm_symbolic_global(
                 rows_b.data_ptr(), ch_ri.data_ptr(), ch_ord.data_ptr(),
                 tot_ch, A_indptr.data_ptr(), A_indices.data_ptr(), bp_ptr,
                 B_indices.data_ptr(), g_keys.data_ptr(), off_b.data_ptr(),
-                sizes_b.data_ptr(), row_nnz.data_ptr(), icode, st)
+                sizes_b.data_ptr(), row_nnz.data_ptr(), icode, ident, st)
         C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
         torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
         nnz = int(C_indptr[-1].item())
@@ -253,7 +264,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     if g_batches:
         g_vals = torch.empty(g_keys.numel(), dtype=A_vals.dtype, device=dev)
         for (rows_b, off_b, sizes_b, total_b, ch_ri, ch_ord,
-             tot_ch) in g_batches:
+             tot_ch, ident) in g_batches:
             g_keys[:total_b].fill_(-1)
             g_vals[:total_b].zero_()
             ext.spgemm_numeric_global_fill(
@@ -261,7 +272,8 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 tot_ch, A_indptr.data_ptr(), A_indices.data_ptr(),
                 A_vals.data_ptr(), bp_ptr, B_indices.data_ptr(),
                 B_vals.data_ptr(), g_keys.data_ptr(), g_vals.data_ptr(),
-                off_b.data_ptr(), sizes_b.data_ptr(), code, icode, st)
+                off_b.data_ptr(), sizes_b.data_ptr(), code, icode, ident,
+                st)
             ext.spgemm_global_compact(
                 rows_b.data_ptr(), rows_b.numel(), g_keys.data_ptr(),
                 g_vals.data_ptr(), off_b.data_ptr(), sizes_b.data_ptr(),

@@ -244,7 +244,7 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_symbolic_global_kernel(
     const I* __restrict__ A_indices, const i64* __restrict__ B_indptr,
     const I* __restrict__ B_indices, uint32_t* __restrict__ tbl_ws,
     const i64* __restrict__ tbl_off, const i64* __restrict__ tbl_size,
-    i64* __restrict__ row_nnz) {
+    i64* __restrict__ row_nnz, int identity) {
   const i64 ch = blockIdx.x;
   if (ch >= n_chunks) return;
   const i64 idx = chunk_rowidx[ch];
@@ -260,7 +260,16 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_symbolic_global_kernel(
   for (i64 jp = a0 + threadIdx.x; jp < a1; jp += blockDim.x) {
     const i64 k = (i64)A_indices[jp];
     for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-      if (tbl_insert(tbl, mask, (uint32_t)B_indices[kp])) ++local;
+      const uint32_t col = (uint32_t)B_indices[kp];
+      if (identity) {
+        // dense-slot mode: table spans all columns, slot = col, the CAS
+        // succeeds or finds col — no probe walk, no cross-slot conflicts
+        if (tbl[col] != col &&
+            atomicCAS(&tbl[col], EMPTY, col) == EMPTY)
+          ++local;
+      } else if (tbl_insert(tbl, mask, col)) {
+        ++local;
+      }
     }
   }
   atomicAdd(&cnt, local);
@@ -476,7 +485,8 @@ void ls_spgemm_symbolic_global(uintptr_t rows_list, uintptr_t chunk_rowidx,
                                uintptr_t B_indptr, uintptr_t B_indices,
                                uintptr_t tbl_ws, uintptr_t tbl_off,
                                uintptr_t tbl_size, uintptr_t row_nnz,
-                               int idx_dtype, uintptr_t stream) {
+                               int idx_dtype, int identity,
+                               uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
       (spgemm_symbolic_global_kernel<idx_t>), dim3(n_chunks),
@@ -490,7 +500,7 @@ void ls_spgemm_symbolic_global(uintptr_t rows_list, uintptr_t chunk_rowidx,
       reinterpret_cast<uint32_t*>(tbl_ws),
       reinterpret_cast<const i64*>(tbl_off),
       reinterpret_cast<const i64*>(tbl_size),
-      reinterpret_cast<i64*>(row_nnz)));
+      reinterpret_cast<i64*>(row_nnz), identity));
   ls_check(hipGetLastError(), "spgemm_symbolic_global");
 }
 
@@ -502,7 +512,7 @@ void ls_spgemm_numeric_global_fill(uintptr_t rows_list,
                                    uintptr_t B_indices, uintptr_t B_vals,
                                    uintptr_t key_ws, uintptr_t val_ws,
                                    uintptr_t tbl_off, uintptr_t tbl_size,
-                                   int dtype, int idx_dtype,
+                                   int dtype, int idx_dtype, int identity,
                                    uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
@@ -519,7 +529,7 @@ void ls_spgemm_numeric_global_fill(uintptr_t rows_list,
       reinterpret_cast<uint32_t*>(key_ws),
       reinterpret_cast<val_t*>(val_ws),
       reinterpret_cast<const i64*>(tbl_off),
-      reinterpret_cast<const i64*>(tbl_size))));
+      reinterpret_cast<const i64*>(tbl_size), identity)));
   ls_check(hipGetLastError(), "spgemm_numeric_global_fill");
 }
 
