@@ -288,7 +288,7 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_fill(
     const i64* __restrict__ B_indptr, const I* __restrict__ B_indices,
     const T* __restrict__ B_vals, uint32_t* __restrict__ key_ws,
     T* __restrict__ val_ws, const i64* __restrict__ tbl_off,
-    const i64* __restrict__ tbl_size) {
+    const i64* __restrict__ tbl_size, int identity) {
   const i64 ch = blockIdx.x;
   if (ch >= n_chunks) return;
   const i64 idx = chunk_rowidx[ch];
@@ -302,8 +302,14 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_fill(
     const i64 k = (i64)A_indices[jp];
     const T a = A_vals[jp];
     for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-      const uint32_t slot = tbl_find_or_insert(
-          keys, mask, (uint32_t)B_indices[kp]);
+      const uint32_t col = (uint32_t)B_indices[kp];
+      uint32_t slot;
+      if (identity) {
+        slot = col;  // dense-slot mode: no probing
+        if (keys[col] != col) keys[col] = col;  // idempotent mark
+      } else {
+        slot = tbl_find_or_insert(keys, mask, col);
+      }
       atomic_add_val(&vals[slot], a * B_vals[kp]);
     }
   }
