@@ -142,12 +142,27 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
       fast:            allocate by upper bound (memory-hungry), numeric
                        counts exact nnz, compact afterwards — skips the
                        whole symbolic pass."""
+    import os as _os
+    import time as _time
+    _timing = _os.environ.get("LS_SPGEMM_TIMING") == "1"
+
+    def _tick(label, _last=[None]):
+        if not _timing:
+            return
+        torch.cuda.synchronize()
+        now = _time.perf_counter()
+        if _last[0] is not None:
+            print(f"  [spgemm] {label}: {(now - _last[0])*1e3:.1f} ms",
+                  flush=True)
+        _last[0] = now
+
     ext = _cext.require_hip()
     dev = A_vals.device
     st = _stream()
     n_rows = A_indptr.numel() - 1
     code = _code(A_vals)
     icode = _icode(A_indices)
+    _tick(None)
     assert A_indices.dtype == B_indices.dtype
     if n_colsB >= 2 ** 32:
         raise NotImplementedError("SpGEMM requires B.shape[1] < 2^32")
@@ -158,6 +173,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     ub = torch.empty(n_rows, dtype=torch.int64, device=dev)
     ext.spgemm_row_ub(A_indptr.data_ptr(), A_indices.data_ptr(),
                       bp_ptr, ub.data_ptr(), n_rows, icode, st)
+    _tick("row_ub")
     cuts = ext.spgemm_lds_bins  # (48, 128, 1024, 4096)
     bins = []
     prev = -1
@@ -217,6 +233,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                                   ch_ord.contiguous(), tot_ch,
                                   1 if ident else 0))
         g_keys = torch.empty(max_total, dtype=torch.int32, device=dev)
+    _tick("binning")
 
     if not fast:
         # ---- exact 2-phase: symbolic then numeric --------------------
@@ -234,6 +251,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 tot_ch, A_indptr.data_ptr(), A_indices.data_ptr(), bp_ptr,
                 B_indices.data_ptr(), g_keys.data_ptr(), off_b.data_ptr(),
                 sizes_b.data_ptr(), row_nnz.data_ptr(), icode, ident, st)
+        _tick("symbolic")
         C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
         torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
         nnz = int(C_indptr[-1].item())
@@ -241,6 +259,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         C_vals = torch.empty(nnz, dtype=A_vals.dtype, device=dev)
         out_indptr = C_indptr
         nnz_ptr = 0
+        _tick("alloc")
     else:
         # ---- fast: allocate by upper bound, numeric counts -----------
         # (clamped: a row has at most n_colsB distinct columns)
@@ -280,6 +299,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 out_indptr.data_ptr(), C_indices.data_ptr(),
                 C_vals.data_ptr(), nnz_ptr, code, icode, st)
 
+    _tick("numeric")
     if fast:
         # compact the capacity layout to exact CSR
         C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
@@ -307,6 +327,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
             order = torch.argsort(keys)
             C_indices[ent] = C_indices[ent][order]
             C_vals[ent] = C_vals[ent][order]
+    _tick("postsort")
     return C_indptr, C_indices, C_vals
 
 
