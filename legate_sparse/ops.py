@@ -175,12 +175,17 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                       bp_ptr, ub.data_ptr(), n_rows, icode, st)
     _tick("row_ub")
     cuts = ext.spgemm_lds_bins  # (48, 128, 1024, 4096)
+    # one-pass binning: bucketize + argsort + a single host transfer
+    edges = torch.tensor(list(cuts), dtype=torch.int64, device=dev)
+    bin_id = torch.bucketize(ub, edges, right=False)  # 0..len(cuts)
+    order = torch.argsort(bin_id, stable=True)
+    counts = torch.bincount(bin_id, minlength=len(cuts) + 1).cpu().tolist()
     bins = []
-    prev = -1
-    for c in cuts:
-        bins.append(torch.nonzero((ub > prev) & (ub <= c)).reshape(-1))
-        prev = c
-    rows_g = torch.nonzero(ub > cuts[-1]).reshape(-1)
+    off = 0
+    for c in counts[:len(cuts)]:
+        bins.append(order[off:off + c].contiguous())
+        off += c
+    rows_g = order[off:].contiguous()
 
     row_nnz = torch.zeros(n_rows, dtype=torch.int64, device=dev)
     g_batches = []
