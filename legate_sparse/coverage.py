@@ -18,7 +18,9 @@ import torch
 _PROFILE_RANGES = False  # flipped by utils.enable_profiler_ranges()
 
 
-def track_provenance(fn, name: str = None):
+def track_provenance(fn=None, name: str = None):
+    if fn is None:
+        return lambda f: track_provenance(f, name)
     label = name or getattr(fn, "__qualname__", str(fn))
 
     @functools.wraps(fn)

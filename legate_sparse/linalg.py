@@ -153,6 +153,10 @@ def cg_axpby(y: torch.Tensor, x: torch.Tensor, a: torch.Tensor,
 # ---------------------------------------------------------------------------
 # CG (reference linalg.py:465-535, CuPy-derived)
 # ---------------------------------------------------------------------------
+from .coverage import track_provenance
+
+
+@track_provenance
 def cg(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
        maxiter: Optional[int] = None, M=None,
        callback: Optional[Callable] = None, conv_test_iters: int = 25):
@@ -222,6 +226,7 @@ def cg(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
 # ---------------------------------------------------------------------------
 # GMRES (reference linalg.py:540-668, restarted, host lstsq)
 # ---------------------------------------------------------------------------
+@track_provenance
 def gmres(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
           restart: Optional[int] = None, maxiter: Optional[int] = None,
           M=None, callback: Optional[Callable] = None):
