@@ -133,7 +133,7 @@ __global__ __launch_bounds__(BLOCK) void spgemm_symbolic_lds_kernel(
 }
 
 template <typename T, typename I, int BLOCK, int ROWS, int TBL,
-          bool COUNT>
+          bool COUNT, int ABL = 0>
 __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
     const i64* __restrict__ rows_list, i64 n_bin_rows,
     const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
@@ -177,11 +177,17 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
             }
             h = (h + 1) & (TBL - 1);
           }
-          atomic_add_val(&vals[group][slot], a * B_vals[kp]);
+          if constexpr (ABL == 2)
+            vals[group][slot] = a * B_vals[kp];
+          else
+            atomic_add_val(&vals[group][slot], a * B_vals[kp]);
         } else {
           const uint32_t slot = tbl_find_or_insert(
               keys[group], TBL - 1, col);
-          atomic_add_val(&vals[group][slot], a * B_vals[kp]);
+          if constexpr (ABL == 2)
+            vals[group][slot] = a * B_vals[kp];
+          else
+            atomic_add_val(&vals[group][slot], a * B_vals[kp]);
         }
       }
     }
@@ -192,7 +198,7 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
   // (0xFFFFFFFF) sorts to the end, so the live entries end up as a sorted
   // prefix.  Compile-time bounds -> uniform barriers.
 #pragma unroll 1
-  for (int k = 2; k <= TBL; k <<= 1) {
+  for (int k = 2; ABL == 0 && k <= TBL; k <<= 1) {
 #pragma unroll 1
     for (int j = k >> 1; j > 0; j >>= 1) {
       for (int t = threadIdx.x % W; t < TBL; t += W) {
@@ -411,6 +417,40 @@ void numeric_lds_dispatch(int cfg, const i64* rl, i64 n, const i64* ap,
   }
 }
 
+template <typename T, typename I, int ABL>
+void numeric_lds_dispatch_abl(int cfg, const i64* rl, i64 n, const i64* ap,
+                              const I* ai, const T* av, const i64* bp,
+                              const I* bi, const T* bv, const i64* cp,
+                              I* ci, T* cv, i64* rn, hipStream_t s) {
+  switch (cfg) {
+    case 0:
+      hipLaunchKernelGGL(
+          (spgemm_numeric_lds_kernel<T, I, 256, 32, 64, false, ABL>),
+          dim3((n + 31) / 32), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi,
+          bv, cp, ci, cv, rn);
+      break;
+    case 1:
+      hipLaunchKernelGGL(
+          (spgemm_numeric_lds_kernel<T, I, 256, 16, 256, false, ABL>),
+          dim3((n + 15) / 16), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi,
+          bv, cp, ci, cv, rn);
+      break;
+    case 2:
+      hipLaunchKernelGGL(
+          (spgemm_numeric_lds_kernel<T, I, 128, 2, 2048, false, ABL>),
+          dim3((n + 1) / 2), dim3(128), 0, s, rl, n, ap, ai, av, bp, bi,
+          bv, cp, ci, cv, rn);
+      break;
+    default: {
+      constexpr int TBL3 = (sizeof(T) > 8) ? 4096 : 8192;
+      hipLaunchKernelGGL(
+          (spgemm_numeric_lds_kernel<T, I, 256, 1, TBL3, false, ABL>),
+          dim3(n), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi, bv, cp, ci,
+          cv, rn);
+    }
+  }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -446,6 +486,8 @@ void ls_spgemm_symbolic_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
   ls_check(hipGetLastError(), "spgemm_symbolic_lds");
 }
 
+#include <cstdlib>
+
 void ls_spgemm_numeric_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
                            uintptr_t A_indptr, uintptr_t A_indices,
                            uintptr_t A_vals, uintptr_t B_indptr,
@@ -454,6 +496,40 @@ void ls_spgemm_numeric_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
                            uintptr_t C_vals, int dtype, int idx_dtype,
                            uintptr_t row_nnz_out, uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  static const char* abl_env = std::getenv("LS_SPGEMM_ABLATE");
+  const int abl = abl_env ? atoi(abl_env) : 0;
+  if (abl == 1 && !row_nnz_out) {
+    DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
+        numeric_lds_dispatch_abl<val_t, idx_t, 1>(
+            cfg, reinterpret_cast<const i64*>(rows_list), n_bin_rows,
+            reinterpret_cast<const i64*>(A_indptr),
+            reinterpret_cast<const idx_t*>(A_indices),
+            reinterpret_cast<const val_t*>(A_vals),
+            reinterpret_cast<const i64*>(B_indptr),
+            reinterpret_cast<const idx_t*>(B_indices),
+            reinterpret_cast<const val_t*>(B_vals),
+            reinterpret_cast<const i64*>(C_indptr),
+            reinterpret_cast<idx_t*>(C_indices),
+            reinterpret_cast<val_t*>(C_vals), nullptr, s))));
+    ls_check(hipGetLastError(), "spgemm_numeric_lds");
+    return;
+  }
+  if (abl == 2 && !row_nnz_out) {
+    DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
+        numeric_lds_dispatch_abl<val_t, idx_t, 2>(
+            cfg, reinterpret_cast<const i64*>(rows_list), n_bin_rows,
+            reinterpret_cast<const i64*>(A_indptr),
+            reinterpret_cast<const idx_t*>(A_indices),
+            reinterpret_cast<const val_t*>(A_vals),
+            reinterpret_cast<const i64*>(B_indptr),
+            reinterpret_cast<const idx_t*>(B_indices),
+            reinterpret_cast<const val_t*>(B_vals),
+            reinterpret_cast<const i64*>(C_indptr),
+            reinterpret_cast<idx_t*>(C_indices),
+            reinterpret_cast<val_t*>(C_vals), nullptr, s))));
+    ls_check(hipGetLastError(), "spgemm_numeric_lds");
+    return;
+  }
   if (row_nnz_out) {
     DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
         numeric_lds_dispatch<val_t, idx_t, true>(
