@@ -216,7 +216,15 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
           }
         }
       }
-      __syncthreads();
+      // each group's table is touched only by its own W lanes; when the
+      // group fits one wave (W <= 64: cfg0-2) lockstep + in-order LDS
+      // makes a block barrier unnecessary — a compiler fence suffices.
+      // Measured: the block barriers were 20.7 of 24.3 ms of the Poisson
+      // numeric phase (LS_SPGEMM_ABLATE).
+      if constexpr (W <= 64)
+        __builtin_amdgcn_wave_barrier();
+      else
+        __syncthreads();
     }
   }
   if (idx < n_bin_rows) {
