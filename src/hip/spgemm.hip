@@ -244,6 +244,95 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
 // per-row power-of-two table region in workspace (keys pre-filled 0xFF,
 // vals pre-zeroed by the host).  Output UNSORTED; host post-sorts these
 // rare rows.
+// Compact-then-sort numeric kernel for wave-contained groups (W <= 64):
+// hash-accumulate into the table, compact live entries to a second LDS
+// region, then bitonic-sort only pow2ceil(cnt) elements (full-table
+// sorting measured 20.7 of 24.3 ms on Poisson — LS_SPGEMM_ABLATE).
+template <typename T, typename I, int BLOCK, int ROWS, int TBL, bool COUNT>
+__global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
+    const i64* __restrict__ rows_list, i64 n_bin_rows,
+    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
+    const T* __restrict__ A_vals, const i64* __restrict__ B_indptr,
+    const I* __restrict__ B_indices, const T* __restrict__ B_vals,
+    const i64* __restrict__ C_indptr, I* __restrict__ C_indices,
+    T* __restrict__ C_vals, i64* __restrict__ row_nnz_out) {
+  constexpr int W = BLOCK / ROWS;
+  static_assert(W <= 64, "groups must be wave-contained");
+  __shared__ uint32_t keys[ROWS][TBL];
+  __shared__ T vals[ROWS][TBL];
+  __shared__ uint32_t okeys[ROWS][TBL];
+  __shared__ T ovals[ROWS][TBL];
+  __shared__ int cur[ROWS];
+  const int group = threadIdx.x / W;
+  const int lane = threadIdx.x % W;
+  const i64 idx = (i64)blockIdx.x * ROWS + group;
+  for (int t = lane; t < TBL; t += W) {
+    keys[group][t] = EMPTY;
+    vals[group][t] = ls_zero<T>();
+  }
+  if (lane == 0) cur[group] = 0;
+  __builtin_amdgcn_wave_barrier();
+  if (idx < n_bin_rows) {
+    const i64 row = rows_list[idx];
+    for (i64 jp = A_indptr[row] + lane; jp < A_indptr[row + 1]; jp += W) {
+      const i64 k = (i64)A_indices[jp];
+      const T a = A_vals[jp];
+      for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
+        const uint32_t slot = tbl_find_or_insert(
+            keys[group], TBL - 1, (uint32_t)B_indices[kp]);
+        atomic_add_val(&vals[group][slot], a * B_vals[kp]);
+      }
+    }
+  }
+  __builtin_amdgcn_wave_barrier();
+  // compact live entries to the front of (okeys, ovals)
+  for (int t = lane; t < TBL; t += W) {
+    const uint32_t c = keys[group][t];
+    if (c != EMPTY) {
+      const int at = atomicAdd(&cur[group], 1);
+      okeys[group][at] = c;
+      ovals[group][at] = vals[group][t];
+    }
+  }
+  __builtin_amdgcn_wave_barrier();
+  const int m = cur[group];
+  int P = 1;
+  while (P < m) P <<= 1;
+  for (int t = lane + m; t < P; t += W) okeys[group][t] = EMPTY;
+  __builtin_amdgcn_wave_barrier();
+#pragma unroll 1
+  for (int k = 2; k <= P; k <<= 1) {
+#pragma unroll 1
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      for (int t = lane; t < P; t += W) {
+        const int tj = t ^ j;
+        if (tj > t) {
+          const bool up = ((t & k) == 0);
+          const uint32_t a = okeys[group][t];
+          const uint32_t b = okeys[group][tj];
+          if ((a > b) == up) {
+            okeys[group][t] = b;
+            okeys[group][tj] = a;
+            const T tv = ovals[group][t];
+            ovals[group][t] = ovals[group][tj];
+            ovals[group][tj] = tv;
+          }
+        }
+      }
+      __builtin_amdgcn_wave_barrier();
+    }
+  }
+  if (idx < n_bin_rows) {
+    const i64 row = rows_list[idx];
+    const i64 base = C_indptr[row];
+    for (int t = lane; t < m; t += W) {
+      C_indices[base + t] = (I)okeys[group][t];
+      C_vals[base + t] = ovals[group][t];
+    }
+    if (COUNT && lane == 0) row_nnz_out[row] = m;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Hub rows are CHUNKED over many workgroups (one workgroup per
 // GLOBAL_CHUNK A-elements of a row) so a power-law row parallelizes
@@ -393,30 +482,45 @@ void numeric_lds_dispatch(int cfg, const i64* rl, i64 n, const i64* ap,
                           const I* ai, const T* av, const i64* bp,
                           const I* bi, const T* bv, const i64* cp, I* ci,
                           T* cv, i64* rn, hipStream_t s) {
+  // cfg0-2 use the compact-then-sort kernel (double LDS buffers: 16-byte
+  // value types shrink ROWS to stay under 160 KiB)
+  constexpr bool big = sizeof(T) > 8;
   switch (cfg) {
     case 0:
       hipLaunchKernelGGL(
-          (spgemm_numeric_lds_kernel<T, I, 256, 32, 64, COUNT>),
+          (spgemm_numeric_lds_csort_kernel<T, I, 256, 32, 64, COUNT>),
           dim3((n + 31) / 32), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi,
           bv, cp, ci, cv, rn);
       break;
     case 1:
-      hipLaunchKernelGGL(
-          (spgemm_numeric_lds_kernel<T, I, 256, 16, 256, COUNT>),
-          dim3((n + 15) / 16), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi,
-          bv, cp, ci, cv, rn);
+      if constexpr (big)
+        hipLaunchKernelGGL(
+            (spgemm_numeric_lds_csort_kernel<T, I, 128, 8, 256, COUNT>),
+            dim3((n + 7) / 8), dim3(128), 0, s, rl, n, ap, ai, av, bp, bi,
+            bv, cp, ci, cv, rn);
+      else
+        hipLaunchKernelGGL(
+            (spgemm_numeric_lds_csort_kernel<T, I, 256, 16, 256, COUNT>),
+            dim3((n + 15) / 16), dim3(256), 0, s, rl, n, ap, ai, av, bp,
+            bi, bv, cp, ci, cv, rn);
       break;
     case 2:
-      hipLaunchKernelGGL(
-          (spgemm_numeric_lds_kernel<T, I, 128, 2, 2048, COUNT>),
-          dim3((n + 1) / 2), dim3(128), 0, s, rl, n, ap, ai, av, bp, bi,
-          bv, cp, ci, cv, rn);
+      if constexpr (big)
+        hipLaunchKernelGGL(
+            (spgemm_numeric_lds_csort_kernel<T, I, 64, 1, 2048, COUNT>),
+            dim3(n), dim3(64), 0, s, rl, n, ap, ai, av, bp, bi, bv, cp,
+            ci, cv, rn);
+      else
+        hipLaunchKernelGGL(
+            (spgemm_numeric_lds_csort_kernel<T, I, 128, 2, 2048, COUNT>),
+            dim3((n + 1) / 2), dim3(128), 0, s, rl, n, ap, ai, av, bp, bi,
+            bv, cp, ci, cv, rn);
       break;
     default: {
       // 16-byte values (c128) at TBL=8192 exceed 160 KiB LDS; use 4096
       // slots there (bin cutoff is ub <= 4096: the table can be exactly
       // full — linear probing still terminates).
-      constexpr int TBL3 = (sizeof(T) > 8) ? 4096 : 8192;
+      constexpr int TBL3 = big ? 4096 : 8192;
       hipLaunchKernelGGL(
           (spgemm_numeric_lds_kernel<T, I, 256, 1, TBL3, COUNT>),
           dim3(n), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi, bv, cp, ci,
