@@ -15,7 +15,8 @@ import scipy.sparse as _sp
 
 from .module import (  # noqa: F401
     csr_array, csr_matrix, dia_array, dia_matrix, diags, eye, identity,
-    mmread, mmwrite, spmv, spgemm_csr_csr_csr, is_sparse_matrix, issparse,
+    mmread, mmwrite, save_npz, load_npz, spmv, spgemm_csr_csr_csr,
+    is_sparse_matrix, issparse,
     isspmatrix, isspmatrix_csr, isspmatrix_dia, coord_ty, nnz_ty,
 )
 from . import linalg  # noqa: F401

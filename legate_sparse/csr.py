@@ -562,6 +562,30 @@ class csr_array(CompressedBase):
     def tocsr(self, copy: bool = False) -> "csr_array":
         return self.copy() if copy else self
 
+    def getnnz(self, axis=None):
+        if axis is None:
+            return self.nnz
+        if axis in (1, -1):
+            return self._indptr[1:] - self._indptr[:-1]
+        if axis == 0:
+            out = torch.zeros(self._shape[1], dtype=torch.int64,
+                              device=self._data.device)
+            out.scatter_add_(0, self._indices.long(),
+                             torch.ones_like(self._indices, dtype=torch.int64))
+            if runtime.world_size > 1:
+                comm.allreduce_(out)
+            return out
+        raise ValueError("axis must be None, 0, or 1")
+
+    def count_nonzero(self) -> int:
+        local = int((self._data != 0).sum())
+        if runtime.world_size > 1:
+            return int(comm.allgather_scalars(local).sum())
+        return local
+
+    def get_shape(self):
+        return self._shape
+
     def tocoo(self):
         """Local COO triple (rows are GLOBAL row ids)."""
         dev = self._data.device

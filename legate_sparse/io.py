@@ -76,6 +76,54 @@ def mmread(source) -> csr_array:
     return csr_array((vals, (rows, cols)), shape=(m, n))
 
 
+def save_npz(file, matrix, compressed: bool = True) -> None:
+    """scipy.sparse.save_npz-compatible checkpoint (the reference has no
+    persistence beyond mmread; this adds CSR checkpoint/restore).  In
+    SPMD mode every rank writes its row block to ``<file>.rank<r>``."""
+    from .runtime import runtime
+
+    A = matrix.tocsr() if hasattr(matrix, "tocsr") else matrix
+    path = str(file)
+    if runtime.world_size > 1:
+        path = f"{path}.rank{runtime.rank}"
+    save = np.savez_compressed if compressed else np.savez
+    save(path,
+         format=np.array("csr", dtype=object),
+         shape=np.asarray(A.shape, dtype=np.int64),
+         row_range=np.asarray(A.row_range, dtype=np.int64),
+         world_size=np.asarray([runtime.world_size]),
+         indptr=A.indptr.cpu().numpy(),
+         indices=A.indices.cpu().numpy(),
+         data=A.data.cpu().numpy())
+
+
+def load_npz(file):
+    """Restore a csr_array written by save_npz (matching world size)."""
+    import torch
+
+    from .csr import csr_array
+    from .runtime import runtime
+
+    path = str(file)
+    if runtime.world_size > 1:
+        path = f"{path}.rank{runtime.rank}"
+    if not path.endswith(".npz"):
+        path += ".npz"
+    with np.load(path, allow_pickle=True) as z:
+        ws = int(z["world_size"][0])
+        if ws != runtime.world_size:
+            raise ValueError(
+                f"checkpoint written at world_size={ws}, "
+                f"loading at {runtime.world_size}")
+        shape = tuple(int(v) for v in z["shape"])
+        dev = runtime.device
+        return csr_array(
+            None, shape=shape,
+            _local=(torch.from_numpy(z["indptr"]).to(dev),
+                    torch.from_numpy(z["indices"]).to(dev),
+                    torch.from_numpy(z["data"]).to(dev)))
+
+
 def mmwrite(target, a) -> None:
     """Minimal MatrixMarket writer (the reference has none; provided for
     round-tripping).  Gathers nothing: rank 0 writes only at world_size==1."""

@@ -8,6 +8,7 @@ from .csr import spmv as _spmv, spgemm_csr_csr_csr as _spgemm
 from .dia import dia_array, dia_matrix  # noqa
 from .gallery import diags as _diags, eye as _eye, identity as _identity
 from .io import mmread as _mmread, mmwrite as _mmwrite
+from .io import save_npz, load_npz  # noqa
 from .types import coord_ty, nnz_ty  # noqa
 
 # provenance labels: rocprof traces attribute kernels to these API entry

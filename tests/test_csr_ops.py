@@ -105,3 +105,15 @@ def test_asformat():
     A = lsp.csr_array(S)
     assert A.asformat("csr") is A
     assert A.asformat(None) is A
+
+
+def test_getnnz_count_nonzero():
+    S = sample_csr(9, 7, 0.4, seed=11)
+    A = lsp.csr_array(S)
+    assert A.getnnz() == S.nnz
+    np.testing.assert_array_equal(to_np(A.getnnz(axis=1)),
+                                  np.diff(S.indptr))
+    np.testing.assert_array_equal(to_np(A.getnnz(axis=0)),
+                                  S.getnnz(axis=0))
+    assert A.count_nonzero() == S.count_nonzero()
+    assert A.get_shape() == S.shape

@@ -249,3 +249,29 @@ def test_gpu_rmat_spgemm():
     C = A @ A
     np.testing.assert_allclose(to_np(C.todense()), (S @ S).toarray(),
                                rtol=1e-9, atol=1e-9)
+
+
+@pytest.mark.parametrize("nnz_per_row", [31, 57])
+def test_gpu_spgemm_complex_big_bins(nnz_per_row):
+    """complex128 takes the reduced-ROWS LDS geometries (cfg1/cfg2)."""
+    A = banded_matrix(500, nnz_per_row=nnz_per_row).astype(np.complex128)
+    A = A + 1j * banded_matrix(500, nnz_per_row=nnz_per_row, seed=9)
+    A = A.tocsr()
+    C = lsp.csr_array(A) @ lsp.csr_array(A)
+    np.testing.assert_allclose(to_np(C.todense()), (A @ A).toarray(),
+                               rtol=1e-10, atol=1e-10)
+
+
+def test_gpu_spgemm_fast_vs_exact_identical():
+    import os
+    A = banded_matrix(3000, nnz_per_row=9)
+    La = lsp.csr_array(A)
+    C1 = La @ La
+    os.environ["LS_FAST_SPGEMM"] = "1"
+    try:
+        C2 = La @ La
+    finally:
+        del os.environ["LS_FAST_SPGEMM"]
+    np.testing.assert_array_equal(to_np(C1.indptr), to_np(C2.indptr))
+    np.testing.assert_array_equal(to_np(C1.indices), to_np(C2.indices))
+    np.testing.assert_allclose(to_np(C1.data), to_np(C2.data), rtol=1e-14)

@@ -65,3 +65,14 @@ def test_mmread_bad_banner(tmp_path):
         f.write("not a matrix market file\n1 1 1\n1 1 1.0\n")
     with pytest.raises(ValueError):
         lsp.mmread(path)
+
+
+def test_save_load_npz(tmp_path):
+    import legate_sparse as lsp2
+    S = sample_csr(14, 11, 0.3, seed=6)
+    A = lsp2.csr_array(S)
+    path = str(tmp_path / "ckpt.npz")
+    lsp2.save_npz(path, A)
+    B = lsp2.load_npz(path)
+    assert B.shape == A.shape
+    np.testing.assert_allclose(to_np(B.todense()), S.toarray())
