@@ -120,7 +120,8 @@ def main():
                 "grid": f"{nx}x{ny}",
                 "rows": M,
                 "nnz": nnz,
-                "parallelism": f"row-split dp{n_gpus}, x allgather halo",
+                "parallelism": f"row-split dp{n_gpus}, windowed halo "
+                               "all-to-all + interior/halo overlap",
                 "spgemm_nnz_per_s": round(spgemm_nnz_per_s, 1),
                 "spgemm_ms_per_iter": round(sg_elapsed / sg_iters * 1e3, 3),
                 "spgemm_out_nnz": c_nnz,

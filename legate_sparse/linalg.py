@@ -182,11 +182,17 @@ def cg(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
         x = _to_local_vec(x0, n, dtype, device).clone()
         r = b - Aop.matvec(x)
 
-    z = Mop.matvec(r)
-    if not isinstance(z, torch.Tensor):
-        z = torch.as_tensor(z, device=device)
-    if z.data_ptr() == r.data_ptr():
-        z = z.clone()
+    # Unpreconditioned CG aliases z = r (saves two full vector passes
+    # per iteration vs copying through an identity preconditioner).
+    ident_M = isinstance(Mop, IdentityOperator)
+    if ident_M:
+        z = r
+    else:
+        z = Mop.matvec(r)
+        if not isinstance(z, torch.Tensor):
+            z = torch.as_tensor(z, device=device)
+        if z.data_ptr() == r.data_ptr():
+            z = z.clone()
     p = z.clone()
     q = torch.empty_like(b)
     rho = _gdot(r, z)
@@ -208,9 +214,11 @@ def cg(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
             if bool((rnorm <= atol_t).item()):
                 info = 0
                 break
-        z = Mop.matvec(r, out=z if isinstance(z, torch.Tensor) else None)
-        if not isinstance(z, torch.Tensor):
-            z = torch.as_tensor(z, device=device)
+        if not ident_M:
+            z = Mop.matvec(r, out=z if isinstance(z, torch.Tensor)
+                           else None)
+            if not isinstance(z, torch.Tensor):
+                z = torch.as_tensor(z, device=device)
         rho1 = rho
         rho = _gdot(r, z)
         # beta = rho / rho1 ; p = z + beta p
