@@ -271,6 +271,11 @@ class csr_array(CompressedBase):
         self._ov_cache = None  # halo/interior split caches values
 
     @property
+    def vals(self) -> torch.Tensor:
+        """Alias of ``data`` (reference API exposes ``.vals``)."""
+        return self._data
+
+    @property
     def indices(self) -> torch.Tensor:
         return self._indices
 
