@@ -1,0 +1,74 @@
+# SPDX-License-Identifier: Apache-2.0
+"""Property-based randomized checks vs scipy (hypothesis).
+
+The reference relies on a few fixed seeds; these sweep random shapes,
+densities, duplicate patterns, and empty corners.
+"""
+import numpy as np
+import pytest
+import scipy.sparse as sp
+
+import legate_sparse as lsp
+from testutils import to_np
+
+hyp = pytest.importorskip("hypothesis")
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+
+@st.composite
+def coo_inputs(draw):
+    m = draw(st.integers(1, 40))
+    n = draw(st.integers(1, 40))
+    nnz = draw(st.integers(0, 120))
+    seed = draw(st.integers(0, 2 ** 31 - 1))
+    rng = np.random.default_rng(seed)
+    rows = rng.integers(0, m, nnz)
+    cols = rng.integers(0, n, nnz)
+    vals = rng.standard_normal(nnz)
+    return m, n, rows, cols, vals
+
+
+@settings(max_examples=40, deadline=None)
+@given(coo_inputs())
+def test_coo_ctor_matches_scipy(inp):
+    m, n, rows, cols, vals = inp
+    A = lsp.csr_array((vals, (rows, cols)), shape=(m, n))
+    S = sp.csr_array((vals, (rows, cols)), shape=(m, n))
+    np.testing.assert_allclose(to_np(A.todense()), S.toarray(),
+                               rtol=1e-12, atol=1e-12)
+    assert A.nnz == S.nnz
+
+
+@settings(max_examples=30, deadline=None)
+@given(coo_inputs(), st.integers(0, 2 ** 31 - 1))
+def test_spmv_matches_scipy(inp, xseed):
+    m, n, rows, cols, vals = inp
+    A = lsp.csr_array((vals, (rows, cols)), shape=(m, n))
+    S = sp.csr_array((vals, (rows, cols)), shape=(m, n))
+    x = np.random.default_rng(xseed).standard_normal(n)
+    np.testing.assert_allclose(to_np(A @ x), S @ x, rtol=1e-10, atol=1e-12)
+
+
+@settings(max_examples=25, deadline=None)
+@given(coo_inputs(), coo_inputs())
+def test_spgemm_matches_scipy(a, b):
+    m, k, rows, cols, vals = a
+    _, n, rows2, cols2, vals2 = b
+    A = lsp.csr_array((vals, (rows, cols)), shape=(m, k))
+    Bs = sp.csr_array((vals2, (rows2 % k if k else rows2, cols2)),
+                      shape=(k, n))
+    B = lsp.csr_array((Bs.data, Bs.indices, Bs.indptr), shape=(k, n))
+    As = sp.csr_array((vals, (rows, cols)), shape=(m, k))
+    C = A @ B
+    np.testing.assert_allclose(to_np(C.todense()), (As @ Bs).toarray(),
+                               rtol=1e-10, atol=1e-10)
+
+
+@settings(max_examples=25, deadline=None)
+@given(coo_inputs())
+def test_transpose_matches_scipy(inp):
+    m, n, rows, cols, vals = inp
+    A = lsp.csr_array((vals, (rows, cols)), shape=(m, n))
+    S = sp.csr_array((vals, (rows, cols)), shape=(m, n))
+    np.testing.assert_allclose(to_np(A.T.todense()), S.T.toarray(),
+                               rtol=1e-12, atol=1e-12)
