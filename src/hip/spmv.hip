@@ -20,6 +20,7 @@
 //     per XCD L2 (x reuse between neighboring rows)
 
 #include "common.h"
+#include <cstdlib>
 
 namespace {
 
@@ -208,10 +209,12 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
       return;
     }
   }
+  static const char* gcap_env = std::getenv("LS_SPMV_GRID");
+  const int gcap = gcap_env ? atoi(gcap_env) : 8192;
   auto launch = [&](auto wtag) {
     constexpr int WS = decltype(wtag)::value;
     constexpr int RPB = LS_THREADS / WS;
-    int grid = grid_1d(n_rows, RPB, 8192);
+    int grid = grid_1d(n_rows, RPB, gcap);
     if constexpr (!is_cplx<T>::value) {
       if (pair) {
         if (swz)
