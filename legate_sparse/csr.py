@@ -279,6 +279,25 @@ class csr_array(CompressedBase):
     def indices(self) -> torch.Tensor:
         return self._indices
 
+    @indices.setter
+    def indices(self, new):
+        t = as_torch_1d(new, device=self._data.device)
+        if t.numel() != self._indices.numel():
+            raise ValueError("indices size mismatch")
+        self._indices = t.to(_index_dtype_for(self._shape)).contiguous()
+        self._halo_cache = None
+        self._ov_cache = None
+
+    @property
+    def has_sorted_indices(self) -> bool:
+        """Column indices are kept sorted within every row by
+        construction (ctors sort; SpGEMM sorts in-kernel)."""
+        return True
+
+    @property
+    def has_canonical_format(self) -> bool:
+        return True
+
     @property
     def indptr(self) -> torch.Tensor:
         return self._indptr
@@ -457,6 +476,14 @@ class csr_array(CompressedBase):
 
     def __matmul__(self, other):
         return self.dot(other)
+
+    def __rmatmul__(self, other):
+        """x @ A == (A.T @ x) for 1-D x (reference csr.py:412-414)."""
+        t = other if isinstance(other, torch.Tensor) else torch.from_numpy(
+            np.ascontiguousarray(other))
+        if t.ndim == 1:
+            return self.transpose().dot(t)
+        raise NotImplementedError("left-multiplication only for vectors")
 
     def matvec(self, x, out=None):
         return self.dot(x, out=out)

@@ -117,3 +117,12 @@ def test_getnnz_count_nonzero():
                                   S.getnnz(axis=0))
     assert A.count_nonzero() == S.count_nonzero()
     assert A.get_shape() == S.shape
+
+
+def test_rmatmul_and_flags():
+    S = sample_csr(10, 8, 0.4, seed=12)
+    A = lsp.csr_array(S)
+    x = np.random.default_rng(13).random(10)
+    np.testing.assert_allclose(to_np(x @ A), x @ S, rtol=1e-12)
+    assert A.has_sorted_indices
+    assert A.has_canonical_format
