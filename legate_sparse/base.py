@@ -83,6 +83,10 @@ class CompressedBase:
         if method == "__call__" and name == "multiply" and len(inputs) == 2:
             other = inputs[0] if inputs[1] is self else inputs[1]
             return self.multiply(other)
+        if method == "__call__" and name == "matmul" and len(inputs) == 2:
+            if inputs[1] is self:           # x @ A
+                return self.__rmatmul__(inputs[0])
+            return self.dot(inputs[1])      # A @ x
         return NotImplemented
 
     def _forward_ufunc(self, name):
