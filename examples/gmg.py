@@ -112,9 +112,11 @@ class GMG(LinearOperator):
         super().__init__(A.shape, dtype=A.dtype)
 
     def _smooth(self, A, dinv, x, b, rounds):
+        from legate_sparse import ops
+        y = torch.empty_like(x)
         for _ in range(rounds):
-            r = b - A @ x
-            x = x + self.omega * dinv * r
+            A.dot(x, out=y)
+            ops.jacobi_update(x, b, y, dinv, self.omega)
         return x
 
     def _vcycle(self, l, b):

@@ -44,6 +44,8 @@ void ls_diagonal(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, i64, int,
 void ls_axpby(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, bool, bool,
               int, uintptr_t);
 void ls_vdot(uintptr_t, uintptr_t, uintptr_t, i64, bool, int, uintptr_t);
+void ls_jacobi(uintptr_t, uintptr_t, uintptr_t, uintptr_t, double, i64,
+               int, uintptr_t);
 
 PYBIND11_MODULE(_hip_kernels, m) {
   m.doc() = "legate_sparse gfx950 HIP kernels";
@@ -61,6 +63,7 @@ PYBIND11_MODULE(_hip_kernels, m) {
   m.def("diagonal", &ls_diagonal);
   m.def("axpby", &ls_axpby);
   m.def("vdot", &ls_vdot);
+  m.def("jacobi", &ls_jacobi);
   m.attr("arch") = "gfx950";
   m.attr("spgemm_lds_bins") = py::make_tuple(48, 128, 1024, 4096);
 }

@@ -35,6 +35,24 @@ __global__ __launch_bounds__(LS_THREADS) void axpby_kernel(
   }
 }
 
+template <typename T>
+__global__ __launch_bounds__(LS_THREADS) void jacobi_kernel(
+    T* __restrict__ x, const T* __restrict__ b, const T* __restrict__ y,
+    const T* __restrict__ dinv, double omega, i64 n) {
+  const i64 stride = (i64)gridDim.x * blockDim.x;
+  for (i64 i = (i64)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if constexpr (is_cplx<T>::value) {
+      T r = b[i] - y[i];
+      T upd = dinv[i] * r;
+      x[i] += T{(decltype(upd.re))(omega) * upd.re,
+                (decltype(upd.re))(omega) * upd.im};
+    } else {
+      x[i] += (T)omega * dinv[i] * (b[i] - y[i]);
+    }
+  }
+}
+
 __device__ inline void atomic_add_out(float* p, float v) { atomicAdd(p, v); }
 __device__ inline void atomic_add_out(double* p, double v) {
   atomicAdd(p, v);
@@ -99,6 +117,18 @@ void ls_axpby(uintptr_t y, uintptr_t x, uintptr_t a, uintptr_t b, i64 n,
     }
   }));
   ls_check(hipGetLastError(), "axpby");
+}
+
+void ls_jacobi(uintptr_t x, uintptr_t b, uintptr_t y, uintptr_t dinv,
+               double omega, i64 n, int dtype, uintptr_t stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  int grid = grid_1d(n, LS_THREADS, 4096);
+  DISPATCH_VAL_T(dtype, hipLaunchKernelGGL(
+      (jacobi_kernel<val_t>), dim3(grid), dim3(LS_THREADS), 0, s,
+      reinterpret_cast<val_t*>(x), reinterpret_cast<const val_t*>(b),
+      reinterpret_cast<const val_t*>(y),
+      reinterpret_cast<const val_t*>(dinv), omega, n));
+  ls_check(hipGetLastError(), "jacobi");
 }
 
 void ls_vdot(uintptr_t x, uintptr_t y, uintptr_t out, i64 n, bool conj,
