@@ -78,3 +78,14 @@ def test_dia_nnz():
     got = lsp.dia_array((data, offsets), shape=(5, 5))
     want = sp.dia_array((data, offsets), shape=(5, 5))
     assert got.nnz == want.nnz
+
+
+def test_dia_ufuncs_and_astype():
+    import numpy as np
+    data = np.abs(np.random.default_rng(3).standard_normal((2, 6))) + 0.1
+    offsets = np.array([0, 1])
+    A = lsp.dia_array((data, offsets), shape=(6, 6))
+    got = A.sqrt().tocsr()
+    want = sp.dia_array((np.sqrt(data), offsets), shape=(6, 6)).tocsr()
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray())
+    assert A.astype(np.float32).dtype == np.dtype(np.float32)

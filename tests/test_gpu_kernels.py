@@ -275,3 +275,19 @@ def test_gpu_spgemm_fast_vs_exact_identical():
     np.testing.assert_array_equal(to_np(C1.indptr), to_np(C2.indptr))
     np.testing.assert_array_equal(to_np(C1.indices), to_np(C2.indices))
     np.testing.assert_allclose(to_np(C1.data), to_np(C2.data), rtol=1e-14)
+
+
+def test_gpu_jacobi_update():
+    from legate_sparse import ops
+    rng = np.random.default_rng(30)
+    n = 50000
+    x0 = rng.standard_normal(n)
+    b = rng.standard_normal(n)
+    y = rng.standard_normal(n)
+    dinv = rng.random(n) + 0.5
+    x = torch.from_numpy(x0.copy()).cuda()
+    ops.jacobi_update(x, torch.from_numpy(b).cuda(),
+                      torch.from_numpy(y).cuda(),
+                      torch.from_numpy(dinv).cuda(), 0.8)
+    want = x0 + 0.8 * dinv * (b - y)
+    np.testing.assert_allclose(to_np(x), want, rtol=1e-13)

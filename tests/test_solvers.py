@@ -121,3 +121,16 @@ def test_sparse_operator_rmatvec():
     v = rng.random(10) + 1j * rng.random(10)
     np.testing.assert_allclose(to_np(op.rmatvec(torch.from_numpy(v))),
                                S.conj().T @ v)
+
+
+def test_gmres_complex():
+    n = 30
+    rng = np.random.default_rng(16)
+    D = (rng.random((n, n)) + 1j * rng.random((n, n)))
+    D = D * (rng.random((n, n)) > 0.6) + n * np.eye(n)
+    S = sp.csr_array(D)
+    b = rng.random(n) + 1j * rng.random(n)
+    x, info = lsp.linalg.gmres(lsp.csr_array(S), b, rtol=1e-10,
+                               restart=20, maxiter=300)
+    assert info == 0
+    assert np.linalg.norm(D @ to_np(x) - b) <= 1e-8 * np.linalg.norm(b)
