@@ -468,16 +468,26 @@ def jacobi_update(x: torch.Tensor, b: torch.Tensor, y: torch.Tensor,
     return x
 
 
-def vdot(x: torch.Tensor, y: torch.Tensor,
-         conj: bool = True) -> torch.Tensor:
+def vdot(x: torch.Tensor, y: torch.Tensor, conj: bool = True,
+         out: torch.Tensor = None) -> torch.Tensor:
     """Local <x, y> as a 1-element device tensor (block-reduce kernel on
-    GPU).  Caller all-reduces across ranks; never .item() in solver loops."""
+    GPU).  Caller all-reduces across ranks; never .item() in solver
+    loops.  ``out`` (zeroed here) enables stable buffers for hipGraph
+    capture."""
     if _use_hip(x):
         ext = _cext.require_hip()
-        out = torch.zeros(1, dtype=x.dtype, device=x.device)
+        if out is None:
+            out = torch.zeros(1, dtype=x.dtype, device=x.device)
+        else:
+            out.zero_()
         ext.vdot(x.data_ptr(), y.data_ptr(), out.data_ptr(), x.numel(),
                  bool(conj), _code(x), _stream())
         return out
     if conj and x.is_complex():
-        return (x.conj() * y).sum().reshape(1)
-    return (x * y).sum().reshape(1)
+        res = (x.conj() * y).sum().reshape(1)
+    else:
+        res = (x * y).sum().reshape(1)
+    if out is not None:
+        out.copy_(res)
+        return out
+    return res
