@@ -205,9 +205,11 @@ def cg(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
     # per-iteration launch/Python overhead (goal: graphs for launch-bound
     # inner loops).  Distributed / preconditioned / callback runs use the
     # eager loop below.
+    import os as _os
     graph = None
     if (ident_M and callback is None and runtime.world_size == 1
-            and device.type == "cuda" and maxiter > 8):
+            and device.type == "cuda" and maxiter > 8
+            and _os.environ.get("LS_CG_GRAPH", "1") != "0"):
         try:
             rho_buf = rho  # 1-element device tensors, stable addresses
             pq_buf = torch.zeros_like(rho_buf)
