@@ -207,9 +207,14 @@ def cg(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
     # eager loop below.
     import os as _os
     graph = None
+    # hipGraph replay has a ~10-16 us floor: a measured win below ~2M
+    # local rows (1.7x at 262K rows), ~2% loss at 16.7M — gate by size
+    # (benchmarks/cg_graph_ab.py)
+    _graph_env = _os.environ.get("LS_CG_GRAPH", "auto")
+    _graph_on = (_graph_env == "1"
+                 or (_graph_env == "auto" and b.numel() <= 2 ** 21))
     if (ident_M and callback is None and runtime.world_size == 1
-            and device.type == "cuda" and maxiter > 8
-            and _os.environ.get("LS_CG_GRAPH", "1") != "0"):
+            and device.type == "cuda" and maxiter > 8 and _graph_on):
         try:
             rho_buf = rho  # 1-element device tensors, stable addresses
             pq_buf = torch.zeros_like(rho_buf)
