@@ -31,3 +31,30 @@ def test_distributed_world2_cpu():
 def test_distributed_world3_cpu():
     # odd world size exercises the unequal tail shard in allgatherv
     _run_dist(3)
+
+
+def _run_example_dist(nproc, script, *args, timeout=420):
+    env = dict(os.environ)
+    env.setdefault("MASTER_ADDR", "127.0.0.1")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        f"--nproc-per-node={nproc}", "--master-addr", "127.0.0.1",
+        "--master-port", "29773",
+        os.path.join(repo, "examples", script), *args,
+    ]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=timeout,
+                       env=env, cwd=repo)
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    return r.stdout
+
+
+def test_pde_distributed_world2():
+    out = _run_example_dist(2, "pde.py", "-n", "64")
+    assert "PASS" in out
+
+
+def test_gmg_distributed_world2():
+    # exercises distributed SpGEMM (R A P), transpose, diagonal, CG+V-cycle
+    out = _run_example_dist(2, "gmg.py", "-k", "5")
+    assert "PASS" in out
