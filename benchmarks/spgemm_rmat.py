@@ -35,8 +35,9 @@ def main():
     A = rmat(args.scale, args.edge_factor, seed=1)
     sync()
     t_gen = time.perf_counter() - t0
+    a_nnz = A.nnz  # collective: must run on every rank
     if runtime.rank == 0:
-        print(f"rmat scale={args.scale}: n={A.shape[0]}, nnz={A.nnz}, "
+        print(f"rmat scale={args.scale}: n={A.shape[0]}, nnz={a_nnz}, "
               f"gen {t_gen*1e3:.0f} ms, ranks={runtime.world_size}",
               flush=True)
     C = A @ A  # warmup

@@ -248,6 +248,9 @@ class csr_array(CompressedBase):
 
     @property
     def nnz(self) -> int:
+        """GLOBAL nnz.  First access is a collective (all-gather of local
+        counts) — in SPMD code call it on every rank, never inside a
+        rank-guarded branch."""
         if self._global_nnz is None:
             local = self._data.numel()
             if runtime.world_size > 1:
