@@ -101,6 +101,21 @@ def main():
         sg_elapsed = float(t.item())
     spgemm_nnz_per_s = c_nnz * sg_iters / sg_elapsed
 
+    # ---- CG (solver pipeline: SpMV + dots + fused axpby) ----
+    cg_iters = 50
+    lsp.linalg.cg(A, x, rtol=0.0, maxiter=10, conv_test_iters=0)  # warmup
+    sync()
+    t0 = time.perf_counter()
+    lsp.linalg.cg(A, x, rtol=0.0, maxiter=cg_iters, conv_test_iters=0)
+    sync()
+    cg_elapsed = time.perf_counter() - t0
+    if ws > 1:
+        t = torch.tensor([cg_elapsed], dtype=torch.float64,
+                         device=runtime.device if on_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        cg_elapsed = float(t.item())
+    cg_ms = cg_elapsed / cg_iters * 1e3
+
     if rank == 0:
         out = {
             "metric": "SpMV GFLOP/s on 5-pt Poisson CSR (fp64)",
@@ -125,6 +140,9 @@ def main():
                 "spgemm_nnz_per_s": round(spgemm_nnz_per_s, 1),
                 "spgemm_ms_per_iter": round(sg_elapsed / sg_iters * 1e3, 3),
                 "spgemm_out_nnz": c_nnz,
+                "cg_ms_per_iter": round(cg_ms, 4),
+                "cg_gflops": round((2 * nnz + 10 * M) / (cg_elapsed /
+                                                         cg_iters) / 1e9, 1),
                 "device": "cuda" if on_gpu else "cpu",
             },
         }
