@@ -248,7 +248,8 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
 // hash-accumulate into the table, compact live entries to a second LDS
 // region, then bitonic-sort only pow2ceil(cnt) elements (full-table
 // sorting measured 20.7 of 24.3 ms on Poisson — LS_SPGEMM_ABLATE).
-template <typename T, typename I, int BLOCK, int ROWS, int TBL, bool COUNT>
+template <typename T, typename I, int BLOCK, int ROWS, int TBL, bool COUNT,
+          int ABL = 0>
 __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
     const i64* __restrict__ rows_list, i64 n_bin_rows,
     const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
@@ -285,6 +286,10 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
     }
   }
   __builtin_amdgcn_wave_barrier();
+  if constexpr (ABL == 3) {  // ablation: keep table live, skip the rest
+    if (lane == 0 && keys[group][0] == 1234567u) C_vals[0] = vals[group][0];
+    return;
+  }
   // compact live entries to the front of (okeys, ovals)
   for (int t = lane; t < TBL; t += W) {
     const uint32_t c = keys[group][t];
@@ -295,6 +300,10 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
     }
   }
   __builtin_amdgcn_wave_barrier();
+  if constexpr (ABL == 4) {
+    if (lane == 0 && okeys[group][0] == 1234567u) C_vals[0] = ovals[group][0];
+    return;
+  }
   const int m = cur[group];
   int P = 1;
   while (P < m) P <<= 1;
@@ -321,6 +330,10 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
       }
       __builtin_amdgcn_wave_barrier();
     }
+  }
+  if constexpr (ABL == 5) {
+    if (lane == 0 && okeys[group][0] == 1234567u) C_vals[0] = ovals[group][0];
+    return;
   }
   if (idx < n_bin_rows) {
     const i64 row = rows_list[idx];
@@ -534,6 +547,14 @@ void numeric_lds_dispatch_abl(int cfg, const i64* rl, i64 n, const i64* ap,
                               const I* ai, const T* av, const i64* bp,
                               const I* bi, const T* bv, const i64* cp,
                               I* ci, T* cv, i64* rn, hipStream_t s) {
+  if constexpr (ABL >= 3) {
+    // csort-kernel phase ablation (cfg0 geometry only; timing probes)
+    hipLaunchKernelGGL(
+        (spgemm_numeric_lds_csort_kernel<T, I, 256, 32, 64, false, ABL>),
+        dim3((n + 31) / 32), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi,
+        bv, cp, ci, cv, rn);
+    return;
+  }
   switch (cfg) {
     case 0:
       hipLaunchKernelGGL(
@@ -610,6 +631,28 @@ void ls_spgemm_numeric_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   static const char* abl_env = std::getenv("LS_SPGEMM_ABLATE");
   const int abl = abl_env ? atoi(abl_env) : 0;
+  if (abl >= 3 && !row_nnz_out) {
+    auto run = [&](auto tag) {
+      constexpr int A = decltype(tag)::value;
+      DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
+          numeric_lds_dispatch_abl<val_t, idx_t, A>(
+              cfg, reinterpret_cast<const i64*>(rows_list), n_bin_rows,
+              reinterpret_cast<const i64*>(A_indptr),
+              reinterpret_cast<const idx_t*>(A_indices),
+              reinterpret_cast<const val_t*>(A_vals),
+              reinterpret_cast<const i64*>(B_indptr),
+              reinterpret_cast<const idx_t*>(B_indices),
+              reinterpret_cast<const val_t*>(B_vals),
+              reinterpret_cast<const i64*>(C_indptr),
+              reinterpret_cast<idx_t*>(C_indices),
+              reinterpret_cast<val_t*>(C_vals), nullptr, s))));
+    };
+    if (abl == 3) run(std::integral_constant<int, 3>{});
+    else if (abl == 4) run(std::integral_constant<int, 4>{});
+    else run(std::integral_constant<int, 5>{});
+    ls_check(hipGetLastError(), "spgemm_numeric_lds");
+    return;
+  }
   if (abl == 1 && !row_nnz_out) {
     DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
         numeric_lds_dispatch_abl<val_t, idx_t, 1>(
