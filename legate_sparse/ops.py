@@ -169,6 +169,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     if bp_ptr is None:
         bp_ptr = B_indptr.data_ptr()
     fast = settings.fast_spgemm
+    pack = 1 if n_colsB < (1 << 24) - 1 else 0
 
     ub = torch.empty(n_rows, dtype=torch.int64, device=dev)
     ext.spgemm_row_ub(A_indptr.data_ptr(), A_indices.data_ptr(),
@@ -284,7 +285,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 bp_ptr, B_indices.data_ptr(),
                 B_vals.data_ptr(), out_indptr.data_ptr(),
                 C_indices.data_ptr(), C_vals.data_ptr(), code, icode,
-                nnz_ptr, st)
+                nnz_ptr, pack, st)
     if g_batches:
         g_vals = torch.empty(g_keys.numel(), dtype=A_vals.dtype, device=dev)
         for (rows_b, off_b, sizes_b, total_b, ch_ri, ch_ord,

@@ -125,3 +125,12 @@ template <typename T>
 __device__ inline T wave_reduce_sum(T v) {
   return group_reduce_sum<T, WAVE_SIZE>(v);
 }
+
+__device__ inline uint32_t wave_reduce_max_u32(uint32_t v) {
+#pragma unroll
+  for (int off = WAVE_SIZE / 2; off > 0; off >>= 1) {
+    uint32_t o = __shfl_xor(v, off, WAVE_SIZE);
+    v = o > v ? o : v;
+  }
+  return v;
+}

@@ -21,7 +21,7 @@ void ls_spgemm_symbolic_lds(int, uintptr_t, i64, uintptr_t, uintptr_t,
 void ls_spgemm_numeric_lds(int, uintptr_t, i64, uintptr_t, uintptr_t,
                            uintptr_t, uintptr_t, uintptr_t, uintptr_t,
                            uintptr_t, uintptr_t, uintptr_t, int, int,
-                           uintptr_t, uintptr_t);
+                           uintptr_t, int, uintptr_t);
 void ls_spgemm_symbolic_global(uintptr_t, uintptr_t, uintptr_t, i64,
                                uintptr_t, uintptr_t, uintptr_t, uintptr_t,
                                uintptr_t, uintptr_t, uintptr_t, uintptr_t,

@@ -249,7 +249,7 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
 // region, then bitonic-sort only pow2ceil(cnt) elements (full-table
 // sorting measured 20.7 of 24.3 ms on Poisson — LS_SPGEMM_ABLATE).
 template <typename T, typename I, int BLOCK, int ROWS, int TBL, bool COUNT,
-          int ABL = 0>
+          int ABL = 0, bool PACK = false>
 __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
     const i64* __restrict__ rows_list, i64 n_bin_rows,
     const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
@@ -259,6 +259,8 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
     T* __restrict__ C_vals, i64* __restrict__ row_nnz_out) {
   constexpr int W = BLOCK / ROWS;
   static_assert(W <= 64, "groups must be wave-contained");
+  constexpr int PACK_BITS = 8;  // slot index bits; requires TBL <= 256
+  static_assert(!PACK || TBL <= (1 << PACK_BITS), "PACK needs small TBL");
   __shared__ uint32_t keys[ROWS][TBL];
   __shared__ T vals[ROWS][TBL];
   __shared__ uint32_t okeys[ROWS][TBL];
@@ -290,13 +292,20 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
     if (lane == 0 && keys[group][0] == 1234567u) C_vals[0] = vals[group][0];
     return;
   }
-  // compact live entries to the front of (okeys, ovals)
+  // compact live entries to the front of (okeys, ovals).  PACK mode keeps
+  // the value slot index in the key's low bits so the sort permutes one
+  // u32 array instead of moving 8/16-byte values.
   for (int t = lane; t < TBL; t += W) {
     const uint32_t c = keys[group][t];
     if (c != EMPTY) {
       const int at = atomicAdd(&cur[group], 1);
-      okeys[group][at] = c;
-      ovals[group][at] = vals[group][t];
+      if constexpr (PACK) {
+        okeys[group][at] = (c << PACK_BITS) | (uint32_t)at;
+        ovals[group][at] = vals[group][t];
+      } else {
+        okeys[group][at] = c;
+        ovals[group][at] = vals[group][t];
+      }
     }
   }
   __builtin_amdgcn_wave_barrier();
@@ -307,6 +316,9 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
   const int m = cur[group];
   int P = 1;
   while (P < m) P <<= 1;
+  // wave-uniform sort bound: divergent per-group loop bounds churn the
+  // exec mask; sorting every group to the wave's max P is cheaper
+  P = (int)wave_reduce_max_u32((uint32_t)P);
   for (int t = lane + m; t < P; t += W) okeys[group][t] = EMPTY;
   __builtin_amdgcn_wave_barrier();
 #pragma unroll 1
@@ -322,9 +334,11 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
           if ((a > b) == up) {
             okeys[group][t] = b;
             okeys[group][tj] = a;
-            const T tv = ovals[group][t];
-            ovals[group][t] = ovals[group][tj];
-            ovals[group][tj] = tv;
+            if constexpr (!PACK) {
+              const T tv = ovals[group][t];
+              ovals[group][t] = ovals[group][tj];
+              ovals[group][tj] = tv;
+            }
           }
         }
       }
@@ -339,8 +353,14 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
     const i64 row = rows_list[idx];
     const i64 base = C_indptr[row];
     for (int t = lane; t < m; t += W) {
-      C_indices[base + t] = (I)okeys[group][t];
-      C_vals[base + t] = ovals[group][t];
+      if constexpr (PACK) {
+        const uint32_t kk = okeys[group][t];
+        C_indices[base + t] = (I)(kk >> PACK_BITS);
+        C_vals[base + t] = ovals[group][kk & (TBL - 1)];
+      } else {
+        C_indices[base + t] = (I)okeys[group][t];
+        C_vals[base + t] = ovals[group][t];
+      }
     }
     if (COUNT && lane == 0) row_nnz_out[row] = m;
   }
@@ -494,28 +514,52 @@ template <typename T, typename I, bool COUNT>
 void numeric_lds_dispatch(int cfg, const i64* rl, i64 n, const i64* ap,
                           const I* ai, const T* av, const i64* bp,
                           const I* bi, const T* bv, const i64* cp, I* ci,
-                          T* cv, i64* rn, hipStream_t s) {
+                          T* cv, i64* rn, bool pack, hipStream_t s) {
   // cfg0-2 use the compact-then-sort kernel (double LDS buffers: 16-byte
-  // value types shrink ROWS to stay under 160 KiB)
+  // value types shrink ROWS to stay under 160 KiB).  pack: sort a single
+  // (col<<8|slot) u32 array instead of moving values (n_cols < 2^24-1).
   constexpr bool big = sizeof(T) > 8;
   switch (cfg) {
     case 0:
-      hipLaunchKernelGGL(
-          (spgemm_numeric_lds_csort_kernel<T, I, 256, 32, 64, COUNT>),
-          dim3((n + 31) / 32), dim3(256), 0, s, rl, n, ap, ai, av, bp, bi,
-          bv, cp, ci, cv, rn);
-      break;
-    case 1:
-      if constexpr (big)
+      if (pack)
         hipLaunchKernelGGL(
-            (spgemm_numeric_lds_csort_kernel<T, I, 128, 8, 256, COUNT>),
-            dim3((n + 7) / 8), dim3(128), 0, s, rl, n, ap, ai, av, bp, bi,
-            bv, cp, ci, cv, rn);
+            (spgemm_numeric_lds_csort_kernel<T, I, 256, 32, 64, COUNT, 0,
+                                             true>),
+            dim3((n + 31) / 32), dim3(256), 0, s, rl, n, ap, ai, av, bp,
+            bi, bv, cp, ci, cv, rn);
       else
         hipLaunchKernelGGL(
-            (spgemm_numeric_lds_csort_kernel<T, I, 256, 16, 256, COUNT>),
-            dim3((n + 15) / 16), dim3(256), 0, s, rl, n, ap, ai, av, bp,
+            (spgemm_numeric_lds_csort_kernel<T, I, 256, 32, 64, COUNT>),
+            dim3((n + 31) / 32), dim3(256), 0, s, rl, n, ap, ai, av, bp,
             bi, bv, cp, ci, cv, rn);
+      break;
+    case 1:
+      if constexpr (big) {
+        if (pack)
+          hipLaunchKernelGGL(
+              (spgemm_numeric_lds_csort_kernel<T, I, 128, 8, 256, COUNT,
+                                               0, true>),
+              dim3((n + 7) / 8), dim3(128), 0, s, rl, n, ap, ai, av, bp,
+              bi, bv, cp, ci, cv, rn);
+        else
+          hipLaunchKernelGGL(
+              (spgemm_numeric_lds_csort_kernel<T, I, 128, 8, 256, COUNT>),
+              dim3((n + 7) / 8), dim3(128), 0, s, rl, n, ap, ai, av, bp,
+              bi, bv, cp, ci, cv, rn);
+      } else {
+        if (pack)
+          hipLaunchKernelGGL(
+              (spgemm_numeric_lds_csort_kernel<T, I, 256, 16, 256, COUNT,
+                                               0, true>),
+              dim3((n + 15) / 16), dim3(256), 0, s, rl, n, ap, ai, av,
+              bp, bi, bv, cp, ci, cv, rn);
+        else
+          hipLaunchKernelGGL(
+              (spgemm_numeric_lds_csort_kernel<T, I, 256, 16, 256,
+                                               COUNT>),
+              dim3((n + 15) / 16), dim3(256), 0, s, rl, n, ap, ai, av,
+              bp, bi, bv, cp, ci, cv, rn);
+      }
       break;
     case 2:
       if constexpr (big)
@@ -627,7 +671,8 @@ void ls_spgemm_numeric_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
                            uintptr_t B_indices, uintptr_t B_vals,
                            uintptr_t C_indptr, uintptr_t C_indices,
                            uintptr_t C_vals, int dtype, int idx_dtype,
-                           uintptr_t row_nnz_out, uintptr_t stream) {
+                           uintptr_t row_nnz_out, int pack,
+                           uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   static const char* abl_env = std::getenv("LS_SPGEMM_ABLATE");
   const int abl = abl_env ? atoi(abl_env) : 0;
@@ -698,7 +743,7 @@ void ls_spgemm_numeric_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
             reinterpret_cast<const i64*>(C_indptr),
             reinterpret_cast<idx_t*>(C_indices),
             reinterpret_cast<val_t*>(C_vals),
-            reinterpret_cast<i64*>(row_nnz_out), s))));
+            reinterpret_cast<i64*>(row_nnz_out), pack != 0, s))));
   } else {
     DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
         numeric_lds_dispatch<val_t, idx_t, false>(
@@ -711,7 +756,7 @@ void ls_spgemm_numeric_lds(int cfg, uintptr_t rows_list, i64 n_bin_rows,
             reinterpret_cast<const val_t*>(B_vals),
             reinterpret_cast<const i64*>(C_indptr),
             reinterpret_cast<idx_t*>(C_indices),
-            reinterpret_cast<val_t*>(C_vals), nullptr, s))));
+            reinterpret_cast<val_t*>(C_vals), nullptr, pack != 0, s))));
   }
   ls_check(hipGetLastError(), "spgemm_numeric_lds");
 }
