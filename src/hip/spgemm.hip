@@ -264,7 +264,9 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
   __shared__ uint32_t keys[ROWS][TBL];
   __shared__ T vals[ROWS][TBL];
   __shared__ uint32_t okeys[ROWS][TBL];
-  __shared__ T ovals[ROWS][TBL];
+  // PACK keeps the TABLE slot in the key's low bits, so no second value
+  // buffer is needed (smaller LDS -> more blocks per CU)
+  __shared__ T ovals[ROWS][PACK ? 1 : TBL];
   __shared__ int cur[ROWS];
   const int group = threadIdx.x / W;
   const int lane = threadIdx.x % W;
@@ -292,16 +294,15 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
     if (lane == 0 && keys[group][0] == 1234567u) C_vals[0] = vals[group][0];
     return;
   }
-  // compact live entries to the front of (okeys, ovals).  PACK mode keeps
-  // the value slot index in the key's low bits so the sort permutes one
-  // u32 array instead of moving 8/16-byte values.
+  // compact live entries to the front of okeys (+ovals without PACK).
+  // PACK keeps the TABLE slot in the key's low bits so the sort permutes
+  // one u32 array and values never move.
   for (int t = lane; t < TBL; t += W) {
     const uint32_t c = keys[group][t];
     if (c != EMPTY) {
       const int at = atomicAdd(&cur[group], 1);
       if constexpr (PACK) {
-        okeys[group][at] = (c << PACK_BITS) | (uint32_t)at;
-        ovals[group][at] = vals[group][t];
+        okeys[group][at] = (c << PACK_BITS) | (uint32_t)t;
       } else {
         okeys[group][at] = c;
         ovals[group][at] = vals[group][t];
@@ -356,7 +357,7 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_csort_kernel(
       if constexpr (PACK) {
         const uint32_t kk = okeys[group][t];
         C_indices[base + t] = (I)(kk >> PACK_BITS);
-        C_vals[base + t] = ovals[group][kk & (TBL - 1)];
+        C_vals[base + t] = vals[group][kk & (TBL - 1)];
       } else {
         C_indices[base + t] = (I)okeys[group][t];
         C_vals[base + t] = ovals[group][t];
