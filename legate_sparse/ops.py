@@ -176,16 +176,26 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                       bp_ptr, ub.data_ptr(), n_rows, icode, st)
     _tick("row_ub")
     cuts = ext.spgemm_lds_bins  # (48, 128, 1024, 4096)
-    # one-pass binning: bucketize + argsort + a single host transfer
+    # one-pass binning: MERGE bins (register k-way merge, rows with few
+    # A-entries and bounded expansion) then LDS-hash bins by expansion
+    # bound, then the HBM bin — one bucketize + argsort + one transfer
+    a_len = A_indptr[1:] - A_indptr[:-1]
+    medges = torch.tensor([8, 32, 64], dtype=torch.int64, device=dev)
+    mcut = torch.bucketize(a_len, medges, right=False)  # 0,1,2 / 3=no
+    mergeable = (mcut < 3) & (ub <= int(cuts[-1]))
     edges = torch.tensor(list(cuts), dtype=torch.int64, device=dev)
-    bin_id = torch.bucketize(ub, edges, right=False)  # 0..len(cuts)
+    bin_id = torch.where(mergeable, mcut,
+                         3 + torch.bucketize(ub, edges, right=False))
     order = torch.argsort(bin_id, stable=True)
-    counts = torch.bincount(bin_id, minlength=len(cuts) + 1).cpu().tolist()
-    bins = []
+    counts = torch.bincount(bin_id,
+                            minlength=len(cuts) + 4).cpu().tolist()
+    groups = []
     off = 0
-    for c in counts[:len(cuts)]:
-        bins.append(order[off:off + c].contiguous())
+    for c in counts[:len(cuts) + 3]:
+        groups.append(order[off:off + c].contiguous())
         off += c
+    mbins = groups[:3]          # merge W=8/32/64
+    bins = groups[3:]           # LDS-hash cfg0..3
     rows_g = order[off:].contiguous()
 
     row_nnz = torch.zeros(n_rows, dtype=torch.int64, device=dev)
@@ -243,6 +253,12 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
 
     if not fast:
         # ---- exact 2-phase: symbolic then numeric --------------------
+        for wcfg, rows in enumerate(mbins):
+            if rows.numel():
+                ext.spgemm_merge_symbolic(
+                    wcfg, rows.data_ptr(), rows.numel(),
+                    A_indptr.data_ptr(), A_indices.data_ptr(), bp_ptr,
+                    B_indices.data_ptr(), row_nnz.data_ptr(), icode, st)
         for cfg, rows in enumerate(bins):
             if rows.numel():
                 ext.spgemm_symbolic_lds(
@@ -277,6 +293,14 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         C_vals = torch.empty(cap, dtype=A_vals.dtype, device=dev)
         nnz_ptr = row_nnz.data_ptr()
 
+    for wcfg, rows in enumerate(mbins):
+        if rows.numel():
+            ext.spgemm_merge_numeric(
+                wcfg, rows.data_ptr(), rows.numel(), A_indptr.data_ptr(),
+                A_indices.data_ptr(), A_vals.data_ptr(), bp_ptr,
+                B_indices.data_ptr(), B_vals.data_ptr(),
+                out_indptr.data_ptr(), C_indices.data_ptr(),
+                C_vals.data_ptr(), code, icode, nnz_ptr, st)
     for cfg, rows in enumerate(bins):
         if rows.numel():
             ext.spgemm_numeric_lds(

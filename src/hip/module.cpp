@@ -15,6 +15,13 @@ void ls_spmv(uintptr_t, uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64,
              i64, int, int, bool, uintptr_t, int, bool, int, int);
 void ls_spgemm_row_ub(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, int,
                       uintptr_t);
+void ls_spgemm_merge_symbolic(int, uintptr_t, i64, uintptr_t, uintptr_t,
+                              uintptr_t, uintptr_t, uintptr_t, int,
+                              uintptr_t);
+void ls_spgemm_merge_numeric(int, uintptr_t, i64, uintptr_t, uintptr_t,
+                             uintptr_t, uintptr_t, uintptr_t, uintptr_t,
+                             uintptr_t, uintptr_t, uintptr_t, int, int,
+                             uintptr_t, uintptr_t);
 void ls_spgemm_symbolic_lds(int, uintptr_t, i64, uintptr_t, uintptr_t,
                             uintptr_t, uintptr_t, uintptr_t, int,
                             uintptr_t);
@@ -51,6 +58,8 @@ PYBIND11_MODULE(_hip_kernels, m) {
   m.doc() = "legate_sparse gfx950 HIP kernels";
   m.def("spmv", &ls_spmv);
   m.def("spgemm_row_ub", &ls_spgemm_row_ub);
+  m.def("spgemm_merge_symbolic", &ls_spgemm_merge_symbolic);
+  m.def("spgemm_merge_numeric", &ls_spgemm_merge_numeric);
   m.def("spgemm_symbolic_lds", &ls_spgemm_symbolic_lds);
   m.def("spgemm_numeric_lds", &ls_spgemm_numeric_lds);
   m.def("spgemm_symbolic_global", &ls_spgemm_symbolic_global);

@@ -244,6 +244,119 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
 // per-row power-of-two table region in workspace (keys pre-filled 0xFF,
 // vals pre-zeroed by the host).  Output UNSORTED; host post-sorts these
 // rare rows.
+// ---------------------------------------------------------------------------
+// Register k-way MERGE path for rows with few A-entries (a_len <= W) and
+// bounded expansion: lane l owns A-entry l and walks its (sorted) B row;
+// each output element is a wave-min over the lanes' cursors plus a masked
+// contribution sum — no LDS, no atomics, no hash, sorted output for free.
+// This covers every row of stencil-type operators (5-pt Poisson: a_len 5).
+// ---------------------------------------------------------------------------
+template <typename I, int W>
+__device__ inline uint32_t grp_min_u32(uint32_t v) {
+#pragma unroll
+  for (int off = W / 2; off > 0; off >>= 1) {
+    const uint32_t o = __shfl_xor(v, off, WAVE_SIZE);
+    v = o < v ? o : v;
+  }
+  return v;
+}
+
+template <typename T, int W>
+__device__ inline T grp_sum(T v) {
+#pragma unroll
+  for (int off = W / 2; off > 0; off >>= 1) {
+    v += ls_shfl_down(v, off);
+  }
+  return v;  // valid in lane 0 of the group
+}
+
+template <typename I, int W>
+__global__ __launch_bounds__(LS_THREADS) void spgemm_merge_symbolic_kernel(
+    const i64* __restrict__ rows_list, i64 n_bin_rows,
+    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
+    const i64* __restrict__ B_indptr, const I* __restrict__ B_indices,
+    i64* __restrict__ row_nnz) {
+  constexpr int ROWS = LS_THREADS / W;
+  const int group = threadIdx.x / W;
+  const int lane = threadIdx.x % W;
+  const i64 idx = (i64)blockIdx.x * ROWS + group;
+  if (idx >= n_bin_rows) return;
+  const i64 row = rows_list[idx];
+  const i64 s = A_indptr[row];
+  const int alen = (int)(A_indptr[row + 1] - s);
+  i64 kp = 0, kend = 0;
+  if (lane < alen) {
+    const i64 k = (i64)A_indices[s + lane];
+    kp = B_indptr[k];
+    kend = B_indptr[k + 1];
+  }
+  uint32_t mycol = (lane < alen && kp < kend) ? (uint32_t)B_indices[kp]
+                                              : EMPTY;
+  i64 cnt = 0;
+  while (true) {
+    const uint32_t mn = grp_min_u32<I, W>(mycol);
+    if (mn == EMPTY) break;
+    ++cnt;
+    if (mycol == mn) {
+      ++kp;
+      mycol = (kp < kend) ? (uint32_t)B_indices[kp] : EMPTY;
+    }
+  }
+  if (lane == 0) row_nnz[row] = cnt;
+}
+
+template <typename T, typename I, int W, bool COUNT>
+__global__ __launch_bounds__(LS_THREADS) void spgemm_merge_numeric_kernel(
+    const i64* __restrict__ rows_list, i64 n_bin_rows,
+    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
+    const T* __restrict__ A_vals, const i64* __restrict__ B_indptr,
+    const I* __restrict__ B_indices, const T* __restrict__ B_vals,
+    const i64* __restrict__ C_indptr, I* __restrict__ C_indices,
+    T* __restrict__ C_vals, i64* __restrict__ row_nnz_out) {
+  constexpr int ROWS = LS_THREADS / W;
+  const int group = threadIdx.x / W;
+  const int lane = threadIdx.x % W;
+  const i64 idx = (i64)blockIdx.x * ROWS + group;
+  if (idx >= n_bin_rows) return;
+  const i64 row = rows_list[idx];
+  const i64 s = A_indptr[row];
+  const int alen = (int)(A_indptr[row + 1] - s);
+  i64 kp = 0, kend = 0;
+  T a = ls_zero<T>();
+  if (lane < alen) {
+    const i64 k = (i64)A_indices[s + lane];
+    a = A_vals[s + lane];
+    kp = B_indptr[k];
+    kend = B_indptr[k + 1];
+  }
+  uint32_t mycol = (lane < alen && kp < kend) ? (uint32_t)B_indices[kp]
+                                              : EMPTY;
+  T myval = (mycol != EMPTY) ? B_vals[kp] : ls_zero<T>();
+  const i64 base = C_indptr[row];
+  i64 out = 0;
+  while (true) {
+    const uint32_t mn = grp_min_u32<I, W>(mycol);
+    if (mn == EMPTY) break;
+    T contrib = (mycol == mn) ? a * myval : ls_zero<T>();
+    contrib = grp_sum<T, W>(contrib);
+    if (lane == 0) {
+      C_indices[base + out] = (I)mn;
+      C_vals[base + out] = contrib;
+    }
+    ++out;
+    if (mycol == mn) {
+      ++kp;
+      if (kp < kend) {
+        mycol = (uint32_t)B_indices[kp];
+        myval = B_vals[kp];
+      } else {
+        mycol = EMPTY;
+      }
+    }
+  }
+  if (COUNT && lane == 0) row_nnz_out[row] = out;
+}
+
 // Compact-then-sort numeric kernel for wave-contained groups (W <= 64):
 // hash-accumulate into the table, compact live entries to a second LDS
 // region, then bitonic-sort only pow2ceil(cnt) elements (full-table
@@ -634,6 +747,96 @@ void numeric_lds_dispatch_abl(int cfg, const i64* rl, i64 n, const i64* ap,
 // ---------------------------------------------------------------------------
 // Host entry points
 // ---------------------------------------------------------------------------
+namespace {
+template <typename I>
+void merge_symbolic_dispatch(int wcfg, const i64* rl, i64 n, const i64* ap,
+                             const I* ai, const i64* bp, const I* bi,
+                             i64* rn, hipStream_t s) {
+  auto L = [&](auto wt) {
+    constexpr int W = decltype(wt)::value;
+    constexpr int ROWS = LS_THREADS / W;
+    hipLaunchKernelGGL((spgemm_merge_symbolic_kernel<I, W>),
+                       dim3((n + ROWS - 1) / ROWS), dim3(LS_THREADS), 0, s,
+                       rl, n, ap, ai, bp, bi, rn);
+  };
+  if (wcfg == 0) L(std::integral_constant<int, 8>{});
+  else if (wcfg == 1) L(std::integral_constant<int, 32>{});
+  else L(std::integral_constant<int, 64>{});
+}
+
+template <typename T, typename I, bool COUNT>
+void merge_numeric_dispatch(int wcfg, const i64* rl, i64 n, const i64* ap,
+                            const I* ai, const T* av, const i64* bp,
+                            const I* bi, const T* bv, const i64* cp, I* ci,
+                            T* cv, i64* rn, hipStream_t s) {
+  auto L = [&](auto wt) {
+    constexpr int W = decltype(wt)::value;
+    constexpr int ROWS = LS_THREADS / W;
+    hipLaunchKernelGGL((spgemm_merge_numeric_kernel<T, I, W, COUNT>),
+                       dim3((n + ROWS - 1) / ROWS), dim3(LS_THREADS), 0, s,
+                       rl, n, ap, ai, av, bp, bi, bv, cp, ci, cv, rn);
+  };
+  if (wcfg == 0) L(std::integral_constant<int, 8>{});
+  else if (wcfg == 1) L(std::integral_constant<int, 32>{});
+  else L(std::integral_constant<int, 64>{});
+}
+}  // namespace
+
+void ls_spgemm_merge_symbolic(int wcfg, uintptr_t rows_list, i64 n,
+                              uintptr_t A_indptr, uintptr_t A_indices,
+                              uintptr_t B_indptr, uintptr_t B_indices,
+                              uintptr_t row_nnz, int idx_dtype,
+                              uintptr_t stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  DISPATCH_IDX_T(idx_dtype, (merge_symbolic_dispatch<idx_t>(
+      wcfg, reinterpret_cast<const i64*>(rows_list), n,
+      reinterpret_cast<const i64*>(A_indptr),
+      reinterpret_cast<const idx_t*>(A_indices),
+      reinterpret_cast<const i64*>(B_indptr),
+      reinterpret_cast<const idx_t*>(B_indices),
+      reinterpret_cast<i64*>(row_nnz), s)));
+  ls_check(hipGetLastError(), "spgemm_merge_symbolic");
+}
+
+void ls_spgemm_merge_numeric(int wcfg, uintptr_t rows_list, i64 n,
+                             uintptr_t A_indptr, uintptr_t A_indices,
+                             uintptr_t A_vals, uintptr_t B_indptr,
+                             uintptr_t B_indices, uintptr_t B_vals,
+                             uintptr_t C_indptr, uintptr_t C_indices,
+                             uintptr_t C_vals, int dtype, int idx_dtype,
+                             uintptr_t row_nnz_out, uintptr_t stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  if (row_nnz_out) {
+    DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
+        merge_numeric_dispatch<val_t, idx_t, true>(
+            wcfg, reinterpret_cast<const i64*>(rows_list), n,
+            reinterpret_cast<const i64*>(A_indptr),
+            reinterpret_cast<const idx_t*>(A_indices),
+            reinterpret_cast<const val_t*>(A_vals),
+            reinterpret_cast<const i64*>(B_indptr),
+            reinterpret_cast<const idx_t*>(B_indices),
+            reinterpret_cast<const val_t*>(B_vals),
+            reinterpret_cast<const i64*>(C_indptr),
+            reinterpret_cast<idx_t*>(C_indices),
+            reinterpret_cast<val_t*>(C_vals),
+            reinterpret_cast<i64*>(row_nnz_out), s))));
+  } else {
+    DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
+        merge_numeric_dispatch<val_t, idx_t, false>(
+            wcfg, reinterpret_cast<const i64*>(rows_list), n,
+            reinterpret_cast<const i64*>(A_indptr),
+            reinterpret_cast<const idx_t*>(A_indices),
+            reinterpret_cast<const val_t*>(A_vals),
+            reinterpret_cast<const i64*>(B_indptr),
+            reinterpret_cast<const idx_t*>(B_indices),
+            reinterpret_cast<const val_t*>(B_vals),
+            reinterpret_cast<const i64*>(C_indptr),
+            reinterpret_cast<idx_t*>(C_indices),
+            reinterpret_cast<val_t*>(C_vals), nullptr, s))));
+  }
+  ls_check(hipGetLastError(), "spgemm_merge_numeric");
+}
+
 void ls_spgemm_row_ub(uintptr_t A_indptr, uintptr_t A_indices,
                       uintptr_t B_indptr, uintptr_t ub, i64 n_rows,
                       int idx_dtype, uintptr_t stream) {
