@@ -32,6 +32,9 @@ def main():
     ap.add_argument("-w", "--warmup", type=int, default=5)
     ap.add_argument("--throughput", action="store_true",
                     help="fixed-iteration throughput mode (no conv test)")
+    ap.add_argument("--exact", action="store_true",
+                    help="manufactured-solution accuracy check "
+                         "(p = sin(pi x) sin(pi y), O(h^2) error)")
     args = ap.parse_args()
 
     nx = ny = args.nx
@@ -42,6 +45,36 @@ def main():
         b = torch.ones(hi - lo, dtype=torch.float64, device=runtime.device)
     print0(f"build: {t_build.elapsed*1e3:.1f} ms  "
            f"(n={n}, nnz={A.nnz}, ranks={runtime.world_size})")
+
+    if args.exact:
+        # solve A u = h^2 f with f = 2 pi^2 sin(pi x) sin(pi y); the
+        # discrete solution converges to p = sin(pi x) sin(pi y) at O(h^2)
+        import math
+        h = 1.0 / (nx + 1)
+        lo, hi = runtime.local_range(n)
+        idx = torch.arange(lo, hi, device=runtime.device)
+        xi = ((idx % nx) + 1).to(torch.float64) * h
+        yi = (torch.div(idx, nx, rounding_mode="floor") + 1).to(
+            torch.float64) * h
+        f = 2 * math.pi ** 2 * torch.sin(math.pi * xi) * \
+            torch.sin(math.pi * yi)
+        p_ref = torch.sin(math.pi * xi) * torch.sin(math.pi * yi)
+        with Timer() as t:
+            u, info = lsp.linalg.cg(A, (h * h) * f, rtol=1e-12,
+                                    maxiter=args.max_iters or 10 * nx)
+        num = torch.sum((u - p_ref) ** 2)
+        den = torch.sum(p_ref ** 2)
+        if runtime.world_size > 1:
+            from legate_sparse import comm
+            comm.allreduce_(num)
+            comm.allreduce_(den)
+        err = float(torch.sqrt(num / den))
+        print0(f"solve: {t.elapsed*1e3:.1f} ms, info={info}, "
+               f"relative L2 error vs exact solution: {err:.3e} "
+               f"(expect O(h^2) ~ {1.0/(nx*nx):.1e})")
+        assert err < 10.0 / (nx * nx), "manufactured-solution check FAILED"
+        print0("PASS")
+        return
 
     if args.throughput:
         iters = args.max_iters or 200

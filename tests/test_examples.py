@@ -79,3 +79,8 @@ def test_gmg_gpu():
         pytest.skip("no GPU")
     out = _run("gmg.py", "-k", "7")
     assert "PASS" in out
+
+
+def test_pde_exact_mode():
+    out = _run("pde.py", "-n", "48", "--exact")
+    assert "PASS" in out
