@@ -21,7 +21,6 @@ SPMD conventions (world_size > 1):
 """
 from __future__ import annotations
 
-import math
 from typing import Optional, Tuple
 
 import numpy as np
@@ -30,12 +29,7 @@ import torch
 from . import comm, ops
 from .base import CompressedBase, as_torch_1d
 from .runtime import runtime
-from .settings import settings
-from .types import (
-    is_supported_value_dtype,
-    to_numpy_dtype,
-    to_torch_dtype,
-)
+from .types import to_numpy_dtype, to_torch_dtype
 
 _INDEX_DTYPE = torch.int64
 

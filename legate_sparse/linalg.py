@@ -16,7 +16,6 @@ All vectors are LOCAL shards of partition(n) in SPMD mode.
 """
 from __future__ import annotations
 
-import math
 from typing import Callable, Optional
 
 import numpy as np
