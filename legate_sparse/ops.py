@@ -199,6 +199,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     mbins = groups[:3]          # merge W=8/32/64
     bins = groups[3:]           # LDS-hash cfg0..3
     rows_g = order[off:].contiguous()
+    _tick("binning.sort")
 
     row_nnz = torch.zeros(n_rows, dtype=torch.int64, device=dev)
     g_batches = []
@@ -250,6 +251,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                                   ch_rowidx.contiguous(),
                                   ch_ord.contiguous(), tot_ch,
                                   1 if ident else 0))
+        _tick("binning.batches")
         g_keys = torch.empty(max_total, dtype=torch.int32, device=dev)
     _tick("binning")
 
