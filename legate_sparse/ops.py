@@ -186,9 +186,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     edges = torch.tensor(list(cuts), dtype=torch.int64, device=dev)
     bin_id = torch.where(mergeable, mcut,
                          3 + torch.bucketize(ub, edges, right=False))
-    # 8 bins fit a byte: radix-sorts one pass instead of eight
-    bin_id = bin_id.to(torch.uint8)
-    order = torch.argsort(bin_id, stable=True)
+    order = torch.argsort(bin_id.to(torch.int32), stable=True)
     counts = torch.bincount(bin_id,
                             minlength=len(cuts) + 4).cpu().tolist()
     groups = []
@@ -222,6 +220,9 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         CH = int(ext.spgemm_global_chunk)
         a_len_g = A_indptr[rows_g + 1] - A_indptr[rows_g]
         max_total = 0
+        # vectorized batch construction: one chunk list per identity
+        # class, one host transfer of the batch boundaries (the previous
+        # per-batch loop of small synced ops cost ~300 ms on R-MAT)
         for ident in (False, True):
             sel = torch.nonzero(ident_all == ident).reshape(-1)
             if not sel.numel():
@@ -230,26 +231,44 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
             csum = torch.cumsum(sizes_s, dim=0)
             batch_id = torch.div(csum - sizes_s, budget,
                                  rounding_mode="floor")
-            for b in range(int(batch_id[-1].item()) + 1):
-                m = batch_id == b
-                rows_b = rows_g[sel[m]]
-                sizes_b = sizes_s[m]
-                off_b = torch.cumsum(sizes_b, dim=0) - sizes_b
-                total_b = int(sizes_b.sum().item())
-                max_total = max(max_total, total_b)
-                # chunk rows' A-ranges so hub rows span many workgroups
-                nch = torch.div(a_len_g[sel[m]] + (CH - 1), CH,
-                                rounding_mode="floor").clamp(min=1)
-                tot_ch = int(nch.sum().item())
-                ch_rowidx = torch.repeat_interleave(
-                    torch.arange(rows_b.numel(), device=dev), nch)
-                ch_start = torch.cumsum(nch, 0) - nch
-                ch_ord = (torch.arange(tot_ch, device=dev)
+            off_global = csum - sizes_s
+            rows_s = rows_g[sel]
+            # chunks over ALL rows of this class (rows sorted by batch)
+            nch = torch.div(a_len_g[sel] + (CH - 1), CH,
+                            rounding_mode="floor").clamp(min=1)
+            ch_start = torch.cumsum(nch, 0) - nch
+            tot_ch_all = int(nch.sum())
+            ch_rowidx_all = torch.repeat_interleave(
+                torch.arange(sel.numel(), device=dev), nch)
+            ch_ord_all = (torch.arange(tot_ch_all, device=dev)
                           - torch.repeat_interleave(ch_start, nch))
-                g_batches.append((rows_b, off_b.contiguous(),
-                                  sizes_b.contiguous(), total_b,
-                                  ch_rowidx.contiguous(),
-                                  ch_ord.contiguous(), tot_ch,
+            n_b = int(batch_id[-1]) + 1
+            bt = torch.arange(n_b, device=dev)
+            row_lo = torch.searchsorted(batch_id, bt, right=False)
+            row_hi = torch.searchsorted(batch_id, bt, right=True)
+            ch_lo = ch_start[row_lo]
+            ch_hi = torch.where(
+                row_hi < sel.numel(), ch_start.take(
+                    torch.clamp(row_hi, max=sel.numel() - 1)),
+                torch.full_like(row_hi, tot_ch_all))
+            ch_hi = torch.where(row_hi < sel.numel(), ch_hi,
+                                torch.full_like(row_hi, tot_ch_all))
+            base = off_global[row_lo]
+            tot = csum[torch.clamp(row_hi - 1, min=0)] - base
+            host = torch.stack([row_lo, row_hi, ch_lo, ch_hi, tot]).cpu()
+            rl, rh, cl, chh, tt = host.tolist()
+            for b in range(n_b):
+                r0, r1 = int(rl[b]), int(rh[b])
+                c0, c1 = int(cl[b]), int(chh[b])
+                rows_b = rows_s[r0:r1]
+                sizes_b = sizes_s[r0:r1].contiguous()
+                off_b = (off_global[r0:r1] - off_global[r0]).contiguous()
+                total_b = int(tt[b]) if r1 > r0 else 0
+                max_total = max(max_total, total_b)
+                ch_rowidx = (ch_rowidx_all[c0:c1] - r0).contiguous()
+                ch_ord = ch_ord_all[c0:c1].contiguous()
+                g_batches.append((rows_b, off_b, sizes_b, total_b,
+                                  ch_rowidx, ch_ord, c1 - c0,
                                   1 if ident else 0))
         _tick("binning.batches")
         g_keys = torch.empty(max_total, dtype=torch.int32, device=dev)
