@@ -197,7 +197,6 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     mbins = groups[:3]          # merge W=8/32/64
     bins = groups[3:]           # LDS-hash cfg0..3
     rows_g = order[off:].contiguous()
-    _tick("binning.sort")
 
     row_nnz = torch.zeros(n_rows, dtype=torch.int64, device=dev)
     g_batches = []
@@ -270,7 +269,6 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 g_batches.append((rows_b, off_b, sizes_b, total_b,
                                   ch_rowidx, ch_ord, c1 - c0,
                                   1 if ident else 0))
-        _tick("binning.batches")
         g_keys = torch.empty(max_total, dtype=torch.int32, device=dev)
     _tick("binning")
 
