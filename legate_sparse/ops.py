@@ -331,6 +331,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 B_vals.data_ptr(), out_indptr.data_ptr(),
                 C_indices.data_ptr(), C_vals.data_ptr(), code, icode,
                 nnz_ptr, pack, st)
+    rows_unsorted = []
     if g_batches:
         g_vals = torch.empty(g_keys.numel(), dtype=A_vals.dtype, device=dev)
         for (rows_b, off_b, sizes_b, total_b, ch_ri, ch_ord,
@@ -344,11 +345,23 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 B_vals.data_ptr(), g_keys.data_ptr(), g_vals.data_ptr(),
                 off_b.data_ptr(), sizes_b.data_ptr(), code, icode, ident,
                 st)
-            ext.spgemm_global_compact(
-                rows_b.data_ptr(), rows_b.numel(), g_keys.data_ptr(),
-                g_vals.data_ptr(), off_b.data_ptr(), sizes_b.data_ptr(),
-                out_indptr.data_ptr(), C_indices.data_ptr(),
-                C_vals.data_ptr(), nnz_ptr, code, icode, st)
+            if ident:
+                # identity tables (slot == col): ordered compaction emits
+                # sorted rows directly
+                ext.spgemm_global_compact_sorted(
+                    rows_b.data_ptr(), rows_b.numel(), g_keys.data_ptr(),
+                    g_vals.data_ptr(), off_b.data_ptr(),
+                    sizes_b.data_ptr(), out_indptr.data_ptr(),
+                    C_indices.data_ptr(), C_vals.data_ptr(), nnz_ptr,
+                    code, icode, st)
+            else:
+                ext.spgemm_global_compact(
+                    rows_b.data_ptr(), rows_b.numel(), g_keys.data_ptr(),
+                    g_vals.data_ptr(), off_b.data_ptr(),
+                    sizes_b.data_ptr(), out_indptr.data_ptr(),
+                    C_indices.data_ptr(), C_vals.data_ptr(), nnz_ptr,
+                    code, icode, st)
+                rows_unsorted.append(rows_b)
 
     _tick("numeric")
     if fast:
@@ -362,22 +375,40 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         C_indices = C_indices[src]
         C_vals = C_vals[src]
 
-    if rows_g.numel():
-        # canonical per-row sort for the (rare) global-bin rows only
-        cnts = (C_indptr[rows_g + 1] - C_indptr[rows_g])
-        total_e = int(cnts.sum().item())
-        if total_e:
-            starts = C_indptr[rows_g]
-            seg = torch.repeat_interleave(
-                torch.arange(rows_g.numel(), device=dev), cnts)
-            seg_start = torch.cumsum(cnts, 0) - cnts
-            pos = torch.arange(total_e, device=dev) - torch.repeat_interleave(
-                seg_start, cnts)
-            ent = torch.repeat_interleave(starts, cnts) + pos
-            keys = seg * int(n_colsB) + C_indices[ent].long()
-            order = torch.argsort(keys)
-            C_indices[ent] = C_indices[ent][order]
-            C_vals[ent] = C_vals[ent][order]
+    if rows_unsorted:
+        # canonical per-row sort for NON-identity global-bin rows, in
+        # entry-bounded batches (an unbatched sort materialized ~3x
+        # total-entries int64 tensors — OOM at R-MAT scale 20)
+        rows_u = torch.cat(rows_unsorted)
+        cnts_all = (C_indptr[rows_u + 1] - C_indptr[rows_u])
+        csum_u = torch.cumsum(cnts_all, 0)
+        budget_e = 1 << 28
+        n_u = rows_u.numel()
+        start_row = 0
+        while start_row < n_u:
+            base_e = int(csum_u[start_row - 1]) if start_row else 0
+            end_row = int(torch.searchsorted(
+                csum_u, torch.tensor(base_e + budget_e, device=dev),
+                right=True))
+            end_row = max(end_row, start_row + 1)
+            end_row = min(end_row, n_u)
+            rg = rows_u[start_row:end_row]
+            cnts = cnts_all[start_row:end_row]
+            total_e = int(cnts.sum())
+            if total_e:
+                starts = C_indptr[rg]
+                seg = torch.repeat_interleave(
+                    torch.arange(rg.numel(), device=dev), cnts)
+                seg_start = torch.cumsum(cnts, 0) - cnts
+                pos = (torch.arange(total_e, device=dev)
+                       - torch.repeat_interleave(seg_start, cnts))
+                ent = torch.repeat_interleave(starts, cnts) + pos
+                keys = seg * int(n_colsB) + C_indices[ent].long()
+                order = torch.argsort(keys)
+                C_indices[ent] = C_indices[ent][order]
+                C_vals[ent] = C_vals[ent][order]
+                del seg, seg_start, pos, ent, keys, order
+            start_row = end_row
     _tick("postsort")
     return C_indptr, C_indices, C_vals
 

@@ -41,6 +41,10 @@ void ls_spgemm_numeric_global_fill(uintptr_t, uintptr_t, uintptr_t, i64,
 void ls_spgemm_global_compact(uintptr_t, i64, uintptr_t, uintptr_t,
                               uintptr_t, uintptr_t, uintptr_t, uintptr_t,
                               uintptr_t, uintptr_t, int, int, uintptr_t);
+void ls_spgemm_global_compact_sorted(uintptr_t, i64, uintptr_t, uintptr_t,
+                                     uintptr_t, uintptr_t, uintptr_t,
+                                     uintptr_t, uintptr_t, uintptr_t, int,
+                                     int, uintptr_t);
 void ls_csr_to_dense(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, i64,
                      int, int, uintptr_t);
 void ls_dense_to_csr_nnz(uintptr_t, uintptr_t, i64, i64, int, uintptr_t);
@@ -65,6 +69,7 @@ PYBIND11_MODULE(_hip_kernels, m) {
   m.def("spgemm_symbolic_global", &ls_spgemm_symbolic_global);
   m.def("spgemm_numeric_global_fill", &ls_spgemm_numeric_global_fill);
   m.def("spgemm_global_compact", &ls_spgemm_global_compact);
+  m.def("spgemm_global_compact_sorted", &ls_spgemm_global_compact_sorted);
   m.attr("spgemm_global_chunk") = 2048;
   m.def("csr_to_dense", &ls_csr_to_dense);
   m.def("dense_to_csr_nnz", &ls_dense_to_csr_nnz);
