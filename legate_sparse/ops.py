@@ -103,7 +103,7 @@ def spgemm_local(
     bp_ptr = B_indptr.data_ptr() - int(b_row_offset) * 8
     if _use_hip(A_vals):
         return _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices,
-                           B_vals, n_colsB, bp_ptr)
+                           B_vals, n_colsB, bp_ptr, b_row_offset)
     if not A_vals.is_cuda and _cext.has_cpu():
         ext = _cext.require_cpu()
         row_nnz = torch.empty(n_rows, dtype=torch.int64)
@@ -129,7 +129,7 @@ def spgemm_local(
 
 
 def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
-                n_colsB, bp_ptr=None):
+                n_colsB, bp_ptr=None, b_row_offset=0):
     """Binned hash-table Gustavson on gfx950 (src/hip/spgemm.hip).
 
     Rows are binned by their expansion upper bound so the per-row hash
@@ -217,11 +217,12 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         ident_all = sizes_all >= p2n
         budget = max(int(sizes_all.max().item()), 1 << 27)  # >= 512 MB keys
         CH = int(ext.spgemm_global_chunk)
-        a_len_g = A_indptr[rows_g + 1] - A_indptr[rows_g]
+        a_len_g2 = A_indptr[rows_g + 1] - A_indptr[rows_g]
+        a_lo_g = A_indptr[rows_g]
         max_total = 0
-        # vectorized batch construction: one chunk list per identity
-        # class, one host transfer of the batch boundaries (the previous
-        # per-batch loop of small synced ops cost ~300 ms on R-MAT)
+        # vectorized batch construction; chunks cover EXPANSION positions
+        # (per-thread binary search over the per-row B-length prefix in
+        # the kernel), so hub rows with few A-entries still parallelize
         for ident in (False, True):
             sel = torch.nonzero(ident_all == ident).reshape(-1)
             if not sel.numel():
@@ -232,8 +233,27 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                                  rounding_mode="floor")
             off_global = csum - sizes_s
             rows_s = rows_g[sel]
-            # chunks over ALL rows of this class (rows sorted by batch)
-            nch = torch.div(a_len_g[sel] + (CH - 1), CH,
+            a_len_s = a_len_g2[sel]
+            ub_s = ub[rows_s]
+            # per-row inclusive B-length prefix over this class's entries
+            tot_e = int(a_len_s.sum())
+            base_ent = torch.cumsum(a_len_s, 0) - a_len_s
+            row_of = torch.repeat_interleave(
+                torch.arange(sel.numel(), device=dev), a_len_s)
+            ent_off = (torch.arange(tot_e, device=dev)
+                       - torch.repeat_interleave(base_ent, a_len_s))
+            a_pos = a_lo_g[sel][row_of] + ent_off
+            k_loc = A_indices[a_pos].long() - b_row_offset
+            blen = B_indptr[k_loc + 1] - B_indptr[k_loc]
+            g = torch.cumsum(blen, 0)
+            rowpre = torch.where(
+                base_ent > 0,
+                g[torch.clamp(base_ent - 1, min=0)],
+                torch.zeros_like(base_ent))
+            blen_prefix = (g - torch.repeat_interleave(rowpre, a_len_s)
+                           ).contiguous()
+            # chunks per row by expansion size
+            nch = torch.div(ub_s + (CH - 1), CH,
                             rounding_mode="floor").clamp(min=1)
             ch_start = torch.cumsum(nch, 0) - nch
             tot_ch_all = int(nch.sum())
@@ -266,9 +286,12 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 max_total = max(max_total, total_b)
                 ch_rowidx = (ch_rowidx_all[c0:c1] - r0).contiguous()
                 ch_ord = ch_ord_all[c0:c1].contiguous()
+                pref_base_b = base_ent[r0:r1].contiguous()
+                a_len_b = a_len_s[r0:r1].contiguous()
                 g_batches.append((rows_b, off_b, sizes_b, total_b,
                                   ch_rowidx, ch_ord, c1 - c0,
-                                  1 if ident else 0))
+                                  1 if ident else 0, blen_prefix,
+                                  pref_base_b, a_len_b))
         g_keys = torch.empty(max_total, dtype=torch.int32, device=dev)
     _tick("binning")
 
@@ -287,13 +310,14 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                     A_indices.data_ptr(), bp_ptr,
                     B_indices.data_ptr(), row_nnz.data_ptr(), icode, st)
         for (rows_b, off_b, sizes_b, total_b, ch_ri, ch_ord,
-             tot_ch, ident) in g_batches:
+             tot_ch, ident, bpre, pbase, albat) in g_batches:
             g_keys[:total_b].fill_(-1)
             ext.spgemm_symbolic_global(
                 rows_b.data_ptr(), ch_ri.data_ptr(), ch_ord.data_ptr(),
                 tot_ch, A_indptr.data_ptr(), A_indices.data_ptr(), bp_ptr,
                 B_indices.data_ptr(), g_keys.data_ptr(), off_b.data_ptr(),
-                sizes_b.data_ptr(), row_nnz.data_ptr(), icode, ident, st)
+                sizes_b.data_ptr(), row_nnz.data_ptr(), icode, ident,
+                bpre.data_ptr(), pbase.data_ptr(), albat.data_ptr(), st)
         _tick("symbolic")
         C_indptr = torch.zeros(n_rows + 1, dtype=torch.int64, device=dev)
         torch.cumsum(row_nnz, dim=0, out=C_indptr[1:])
@@ -335,7 +359,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
     if g_batches:
         g_vals = torch.empty(g_keys.numel(), dtype=A_vals.dtype, device=dev)
         for (rows_b, off_b, sizes_b, total_b, ch_ri, ch_ord,
-             tot_ch, ident) in g_batches:
+             tot_ch, ident, bpre, pbase, albat) in g_batches:
             g_keys[:total_b].fill_(-1)
             g_vals[:total_b].zero_()
             ext.spgemm_numeric_global_fill(
@@ -344,7 +368,7 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                 A_vals.data_ptr(), bp_ptr, B_indices.data_ptr(),
                 B_vals.data_ptr(), g_keys.data_ptr(), g_vals.data_ptr(),
                 off_b.data_ptr(), sizes_b.data_ptr(), code, icode, ident,
-                st)
+                bpre.data_ptr(), pbase.data_ptr(), albat.data_ptr(), st)
             if ident:
                 # identity tables (slot == col): ordered compaction emits
                 # sorted rows directly

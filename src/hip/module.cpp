@@ -32,12 +32,14 @@ void ls_spgemm_numeric_lds(int, uintptr_t, i64, uintptr_t, uintptr_t,
 void ls_spgemm_symbolic_global(uintptr_t, uintptr_t, uintptr_t, i64,
                                uintptr_t, uintptr_t, uintptr_t, uintptr_t,
                                uintptr_t, uintptr_t, uintptr_t, uintptr_t,
-                               int, int, uintptr_t);
+                               int, int, uintptr_t, uintptr_t, uintptr_t,
+                               uintptr_t);
 void ls_spgemm_numeric_global_fill(uintptr_t, uintptr_t, uintptr_t, i64,
                                    uintptr_t, uintptr_t, uintptr_t,
                                    uintptr_t, uintptr_t, uintptr_t,
                                    uintptr_t, uintptr_t, uintptr_t,
-                                   uintptr_t, int, int, int, uintptr_t);
+                                   uintptr_t, int, int, int, uintptr_t,
+                                   uintptr_t, uintptr_t, uintptr_t);
 void ls_spgemm_global_compact(uintptr_t, i64, uintptr_t, uintptr_t,
                               uintptr_t, uintptr_t, uintptr_t, uintptr_t,
                               uintptr_t, uintptr_t, int, int, uintptr_t);

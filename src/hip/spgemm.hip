@@ -494,32 +494,47 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_symbolic_global_kernel(
     const I* __restrict__ A_indices, const i64* __restrict__ B_indptr,
     const I* __restrict__ B_indices, uint32_t* __restrict__ tbl_ws,
     const i64* __restrict__ tbl_off, const i64* __restrict__ tbl_size,
-    i64* __restrict__ row_nnz, int identity) {
+    i64* __restrict__ row_nnz, int identity,
+    const i64* __restrict__ blen_prefix, const i64* __restrict__ pref_base,
+    const i64* __restrict__ a_len_g) {
+  // Work is chunked by EXPANSION POSITION (not A-entry): thread e finds
+  // its (A-entry, B-offset) by binary search in the row's B-length
+  // prefix — hub rows with few A-entries but huge B-rows parallelize
+  // fully instead of serializing a per-lane B walk.
   const i64 ch = blockIdx.x;
   if (ch >= n_chunks) return;
   const i64 idx = chunk_rowidx[ch];
   const i64 row = rows_list[idx];
   uint32_t* tbl = tbl_ws + tbl_off[idx];
   const uint32_t mask = (uint32_t)tbl_size[idx] - 1;
-  const i64 a0 = A_indptr[row] + chunk_ord[ch] * GLOBAL_CHUNK;
-  const i64 a1 = min(a0 + GLOBAL_CHUNK, A_indptr[row + 1]);
+  const i64 pb = pref_base[idx];
+  const i64 alen = a_len_g[idx];
+  const i64 ub_row = alen ? blen_prefix[pb + alen - 1] : 0;
+  const i64 e0 = chunk_ord[ch] * GLOBAL_CHUNK;
+  const i64 e1 = min(e0 + GLOBAL_CHUNK, ub_row);
+  const i64 a_base = A_indptr[row];
   __shared__ int cnt;
   if (threadIdx.x == 0) cnt = 0;
   __syncthreads();
   int local = 0;
-  for (i64 jp = a0 + threadIdx.x; jp < a1; jp += blockDim.x) {
-    const i64 k = (i64)A_indices[jp];
-    for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-      const uint32_t col = (uint32_t)B_indices[kp];
-      if (identity) {
-        // dense-slot mode: table spans all columns, slot = col, the CAS
-        // succeeds or finds col — no probe walk, no cross-slot conflicts
-        if (tbl[col] != col &&
-            atomicCAS(&tbl[col], EMPTY, col) == EMPTY)
-          ++local;
-      } else if (tbl_insert(tbl, mask, col)) {
+  for (i64 e = e0 + threadIdx.x; e < e1; e += blockDim.x) {
+    // smallest i with blen_prefix[pb+i] > e
+    i64 lo = 0, hi = alen - 1;
+    while (lo < hi) {
+      const i64 mid = (lo + hi) >> 1;
+      if (blen_prefix[pb + mid] > e)
+        hi = mid;
+      else
+        lo = mid + 1;
+    }
+    const i64 b_off = e - (lo ? blen_prefix[pb + lo - 1] : 0);
+    const i64 k = (i64)A_indices[a_base + lo];
+    const uint32_t col = (uint32_t)B_indices[B_indptr[k] + b_off];
+    if (identity) {
+      if (tbl[col] != col && atomicCAS(&tbl[col], EMPTY, col) == EMPTY)
         ++local;
-      }
+    } else if (tbl_insert(tbl, mask, col)) {
+      ++local;
     }
   }
   atomicAdd(&cnt, local);
@@ -538,7 +553,9 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_fill(
     const i64* __restrict__ B_indptr, const I* __restrict__ B_indices,
     const T* __restrict__ B_vals, uint32_t* __restrict__ key_ws,
     T* __restrict__ val_ws, const i64* __restrict__ tbl_off,
-    const i64* __restrict__ tbl_size, int identity) {
+    const i64* __restrict__ tbl_size, int identity,
+    const i64* __restrict__ blen_prefix, const i64* __restrict__ pref_base,
+    const i64* __restrict__ a_len_g) {
   const i64 ch = blockIdx.x;
   if (ch >= n_chunks) return;
   const i64 idx = chunk_rowidx[ch];
@@ -546,22 +563,34 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_numeric_global_fill(
   uint32_t* keys = key_ws + tbl_off[idx];
   T* vals = val_ws + tbl_off[idx];
   const uint32_t mask = (uint32_t)tbl_size[idx] - 1;
-  const i64 a0 = A_indptr[row] + chunk_ord[ch] * GLOBAL_CHUNK;
-  const i64 a1 = min(a0 + GLOBAL_CHUNK, A_indptr[row + 1]);
-  for (i64 jp = a0 + threadIdx.x; jp < a1; jp += blockDim.x) {
-    const i64 k = (i64)A_indices[jp];
-    const T a = A_vals[jp];
-    for (i64 kp = B_indptr[k]; kp < B_indptr[k + 1]; ++kp) {
-      const uint32_t col = (uint32_t)B_indices[kp];
-      uint32_t slot;
-      if (identity) {
-        slot = col;  // dense-slot mode: no probing
-        if (keys[col] != col) keys[col] = col;  // idempotent mark
-      } else {
-        slot = tbl_find_or_insert(keys, mask, col);
-      }
-      atomic_add_val(&vals[slot], a * B_vals[kp]);
+  const i64 pb = pref_base[idx];
+  const i64 alen = a_len_g[idx];
+  const i64 ub_row = alen ? blen_prefix[pb + alen - 1] : 0;
+  const i64 e0 = chunk_ord[ch] * GLOBAL_CHUNK;
+  const i64 e1 = min(e0 + GLOBAL_CHUNK, ub_row);
+  const i64 a_base = A_indptr[row];
+  for (i64 e = e0 + threadIdx.x; e < e1; e += blockDim.x) {
+    i64 lo = 0, hi = alen - 1;
+    while (lo < hi) {
+      const i64 mid = (lo + hi) >> 1;
+      if (blen_prefix[pb + mid] > e)
+        hi = mid;
+      else
+        lo = mid + 1;
     }
+    const i64 b_off = e - (lo ? blen_prefix[pb + lo - 1] : 0);
+    const i64 k = (i64)A_indices[a_base + lo];
+    const T a = A_vals[a_base + lo];
+    const i64 bp0 = B_indptr[k] + b_off;
+    const uint32_t col = (uint32_t)B_indices[bp0];
+    uint32_t slot;
+    if (identity) {
+      slot = col;
+      if (keys[col] != col) keys[col] = col;  // idempotent mark
+    } else {
+      slot = tbl_find_or_insert(keys, mask, col);
+    }
+    atomic_add_val(&vals[slot], a * B_vals[bp0]);
   }
 }
 
@@ -1023,7 +1052,8 @@ void ls_spgemm_symbolic_global(uintptr_t rows_list, uintptr_t chunk_rowidx,
                                uintptr_t tbl_ws, uintptr_t tbl_off,
                                uintptr_t tbl_size, uintptr_t row_nnz,
                                int idx_dtype, int identity,
-                               uintptr_t stream) {
+                               uintptr_t blen_prefix, uintptr_t pref_base,
+                               uintptr_t a_len_g, uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
       (spgemm_symbolic_global_kernel<idx_t>), dim3(n_chunks),
@@ -1037,7 +1067,10 @@ void ls_spgemm_symbolic_global(uintptr_t rows_list, uintptr_t chunk_rowidx,
       reinterpret_cast<uint32_t*>(tbl_ws),
       reinterpret_cast<const i64*>(tbl_off),
       reinterpret_cast<const i64*>(tbl_size),
-      reinterpret_cast<i64*>(row_nnz), identity));
+      reinterpret_cast<i64*>(row_nnz), identity,
+      reinterpret_cast<const i64*>(blen_prefix),
+      reinterpret_cast<const i64*>(pref_base),
+      reinterpret_cast<const i64*>(a_len_g)));
   ls_check(hipGetLastError(), "spgemm_symbolic_global");
 }
 
@@ -1050,6 +1083,8 @@ void ls_spgemm_numeric_global_fill(uintptr_t rows_list,
                                    uintptr_t key_ws, uintptr_t val_ws,
                                    uintptr_t tbl_off, uintptr_t tbl_size,
                                    int dtype, int idx_dtype, int identity,
+                                   uintptr_t blen_prefix,
+                                   uintptr_t pref_base, uintptr_t a_len_g,
                                    uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
@@ -1066,7 +1101,10 @@ void ls_spgemm_numeric_global_fill(uintptr_t rows_list,
       reinterpret_cast<uint32_t*>(key_ws),
       reinterpret_cast<val_t*>(val_ws),
       reinterpret_cast<const i64*>(tbl_off),
-      reinterpret_cast<const i64*>(tbl_size), identity)));
+      reinterpret_cast<const i64*>(tbl_size), identity,
+      reinterpret_cast<const i64*>(blen_prefix),
+      reinterpret_cast<const i64*>(pref_base),
+      reinterpret_cast<const i64*>(a_len_g))));
   ls_check(hipGetLastError(), "spgemm_numeric_global_fill");
 }
 
