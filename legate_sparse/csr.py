@@ -406,7 +406,17 @@ class csr_array(CompressedBase):
                      y: Optional[torch.Tensor] = None) -> torch.Tensor:
         """Distributed matvec with comm/compute overlap: the halo
         all-to-all runs async (RCCL stream) while the interior SpMV
-        computes on the compute stream; the halo part then accumulates."""
+        computes on the compute stream; the halo part then accumulates.
+
+        LS_PRECISE_IMAGES=0 falls back to a full x all-gather with no
+        window/overlap (the coarse-image escape hatch; the reference's
+        LEGATE_SPARSE_PRECISE_IMAGES analogue)."""
+        from .settings import settings as _settings
+        if not _settings.precise_images:
+            part_x = runtime.partition(self._shape[1])
+            x_full = comm.allgatherv(x, part_x.counts())
+            return ops.spmv(self._indptr, self._indices, self._data,
+                            x_full, y)
         N = self._shape[1]
         part = runtime.partition(N)
         lo = part.lo(runtime.rank)
