@@ -111,42 +111,79 @@ class GMG(LinearOperator):
             nx = nxc
         super().__init__(A.shape, dtype=A.dtype)
 
-    def _smooth(self, A, dinv, x, b, rounds):
+    def _smooth_into(self, l, rounds):
         from legate_sparse import ops
-        y = torch.empty_like(x)
-        for _ in range(rounds):
-            A.dot(x, out=y)
-            ops.jacobi_update(x, b, y, dinv, self.omega)
-        return x
-
-    def _vcycle(self, l, b):
         A, dinv = self.levels[l]
-        x = torch.zeros_like(b)
-        if l == len(self.levels) - 1:
-            return self._smooth(A, dinv, x, b, self.coarse_iters)
-        x = self._smooth(A, dinv, x, b, self.nu)
-        r = b - A @ x
-        P = self.P[l]
-        rc = self._restrict(l, r)
-        ec = self._vcycle(l + 1, rc)
-        x = x + (P @ ec)
-        x = self._smooth(A, dinv, x, b, self.nu)
-        return x
+        buf = self._buf[l]
+        for _ in range(rounds):
+            A.dot(buf["x"], out=buf["y"])
+            ops.jacobi_update(buf["x"], buf["b"], buf["y"], dinv,
+                              self.omega)
 
-    def _restrict(self, l, r):
-        # R = P^T cached per level
-        if not hasattr(self, "_Rt"):
-            self._Rt = {}
-        if l not in self._Rt:
-            self._Rt[l] = self.P[l].T
-        return self._Rt[l] @ r
+    def _ensure_buffers(self):
+        if hasattr(self, "_buf"):
+            return
+        self._buf = []
+        for A, dinv in self.levels:
+            t = dinv  # local shard prototype
+            self._buf.append({
+                "x": torch.zeros_like(t),
+                "b": torch.zeros_like(t),
+                "y": torch.empty_like(t),
+                "r": torch.empty_like(t),
+            })
+        self._Rt = [P.T for P in self.P]
+
+    def _vcycle_buffers(self):
+        """V-cycle over the preallocated per-level buffers, every op
+        in-place — hipGraph-capturable (fixed addresses, no allocs)."""
+        n_lev = len(self.levels)
+        for l in range(n_lev):
+            buf = self._buf[l]
+            buf["x"].zero_()
+            if l == n_lev - 1:
+                self._smooth_into(l, self.coarse_iters)
+                break
+            self._smooth_into(l, self.nu)
+            A, _ = self.levels[l]
+            A.dot(buf["x"], out=buf["y"])
+            torch.sub(buf["b"], buf["y"], out=buf["r"])
+            self._Rt[l].dot(buf["r"], out=self._buf[l + 1]["b"])
+        for l in range(n_lev - 2, -1, -1):
+            buf = self._buf[l]
+            self.P[l].dot(self._buf[l + 1]["x"], out=buf["y"])
+            buf["x"].add_(buf["y"])
+            self._smooth_into(l, self.nu)
 
     def matvec(self, x, out=None):
-        y = self._vcycle(0, x)
+        from legate_sparse.runtime import runtime as _rt
+        self._ensure_buffers()
+        use_graph = (x.is_cuda and _rt.world_size == 1)
+        if use_graph and getattr(self, "_graph", None) is None:
+            try:
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    self._buf[0]["b"].copy_(x)
+                    self._vcycle_buffers()  # warmup
+                torch.cuda.current_stream().wait_stream(side)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._vcycle_buffers()
+                self._graph = g
+            except Exception:
+                self._graph = False
+        if use_graph and self._graph:
+            self._buf[0]["b"].copy_(x)
+            self._graph.replay()
+        else:
+            self._buf[0]["b"].copy_(x)
+            self._vcycle_buffers()
+        y = self._buf[0]["x"]
         if out is not None:
             out.copy_(y)
             return out
-        return y
+        return y.clone()
 
 
 def main():
