@@ -158,7 +158,9 @@ class GMG(LinearOperator):
     def matvec(self, x, out=None):
         from legate_sparse.runtime import runtime as _rt
         self._ensure_buffers()
-        use_graph = (x.is_cuda and _rt.world_size == 1)
+        import os as _os
+        use_graph = (x.is_cuda and _rt.world_size == 1
+                     and _os.environ.get("LS_GMG_GRAPH", "1") != "0")
         if use_graph and getattr(self, "_graph", None) is None:
             try:
                 side = torch.cuda.Stream()
