@@ -159,8 +159,12 @@ class GMG(LinearOperator):
         from legate_sparse.runtime import runtime as _rt
         self._ensure_buffers()
         import os as _os
+        # graph replay is +17% per V-cycle (1.29 vs 1.54 ms at k=11) but
+        # capture costs ~60 ms — it only amortizes for long solves, so
+        # it is opt-in (LS_GMG_GRAPH=1); a typical GMG solve converges
+        # in ~10 iterations
         use_graph = (x.is_cuda and _rt.world_size == 1
-                     and _os.environ.get("LS_GMG_GRAPH", "1") != "0")
+                     and _os.environ.get("LS_GMG_GRAPH", "0") == "1")
         if use_graph and getattr(self, "_graph", None) is None:
             try:
                 side = torch.cuda.Stream()
