@@ -110,6 +110,7 @@ class GMG(LinearOperator):
             Al = (R @ Al) @ P
             nx = nxc
         super().__init__(A.shape, dtype=A.dtype)
+        self._ensure_buffers()  # per-level buffers + cached R = P^T
 
     def _smooth_into(self, l, rounds):
         from legate_sparse import ops
