@@ -214,6 +214,9 @@ def main():
         print0(f"  level {l}: n={Al.shape[0]}, nnz={Al.nnz}")
 
     b = torch.ones(hi - lo, dtype=torch.float64, device=runtime.device)
+    # warmup: first-dispatch costs (code-object loads, allocator, torch
+    # op init) otherwise land in the timed solve
+    lsp.linalg.cg(A, b, rtol=0.0, maxiter=2, M=M, conv_test_iters=0)
     iters = []
     with Timer() as t:
         x, info = lsp.linalg.cg(A, b, rtol=args.rtol, maxiter=args.maxiter,
