@@ -139,6 +139,41 @@ def aslinearoperator(A):
     raise TypeError(f"cannot wrap {type(A)} as LinearOperator")
 
 
+def norm(A, ord="fro"):
+    """Matrix norm of a sparse array: 'fro', 1 (max column abs sum), or
+    inf (max row abs sum) — scipy.sparse.linalg.norm-compatible subset."""
+    import math as _math
+
+    if not hasattr(A, "indptr"):
+        raise TypeError("norm expects a sparse array")
+    data = A.data
+    absd = data.abs()
+    if ord in ("fro", None):
+        s = (absd * absd).sum().reshape(1)
+        if runtime.world_size > 1:
+            comm.allreduce_(s)
+        return float(torch.sqrt(s))
+    if ord == 1:
+        col = torch.zeros(A.shape[1], dtype=absd.dtype, device=absd.device)
+        col.scatter_add_(0, A.indices.long(), absd)
+        if runtime.world_size > 1:
+            comm.allreduce_(col)
+        return float(col.max()) if col.numel() else 0.0
+    if ord in (np.inf, float("inf"), "inf"):
+        row = torch.zeros(A.indptr.numel() - 1, dtype=absd.dtype,
+                          device=absd.device)
+        ids = torch.repeat_interleave(
+            torch.arange(row.numel(), device=absd.device),
+            A.indptr[1:] - A.indptr[:-1])
+        row.scatter_add_(0, ids, absd)
+        m = row.max().reshape(1) if row.numel() else torch.zeros(
+            1, dtype=absd.dtype, device=absd.device)
+        if runtime.world_size > 1:
+            comm.allreduce_(m, op="max")
+        return float(m)
+    raise NotImplementedError(f"norm ord={ord!r} not supported")
+
+
 # ---------------------------------------------------------------------------
 # Fused CG update (reference linalg.py:433-451 + axpby.cu)
 # ---------------------------------------------------------------------------

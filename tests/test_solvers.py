@@ -134,3 +134,13 @@ def test_gmres_complex():
                                restart=20, maxiter=300)
     assert info == 0
     assert np.linalg.norm(D @ to_np(x) - b) <= 1e-8 * np.linalg.norm(b)
+
+
+def test_sparse_norms():
+    from legate_sparse.linalg import norm
+    import scipy.sparse.linalg as spla
+    S = sample_csr(14, 9, 0.4, seed=20)
+    A = lsp.csr_array(S)
+    assert abs(norm(A) - spla.norm(S)) < 1e-12
+    assert abs(norm(A, 1) - spla.norm(S, 1)) < 1e-12
+    assert abs(norm(A, np.inf) - spla.norm(S, np.inf)) < 1e-12
