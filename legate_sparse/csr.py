@@ -512,6 +512,79 @@ class csr_array(CompressedBase):
 
     __rmul__ = __mul__
 
+    def _add_sparse(self, other: "csr_array", alpha=1.0) -> "csr_array":
+        """C = self + alpha*other (same shape; partitions match by
+        construction).  Local concat -> sort -> duplicate sum."""
+        if self.shape != other.shape:
+            raise ValueError(f"shape mismatch {self.shape} vs {other.shape}")
+        dt = _common_dtype(self._data, other._data)
+        dev = self._data.device
+        lr = self._row_hi - self._row_lo
+        r1 = torch.repeat_interleave(
+            torch.arange(lr, device=dev),
+            self._indptr[1:] - self._indptr[:-1])
+        r2 = torch.repeat_interleave(
+            torch.arange(lr, device=dev),
+            other._indptr[1:] - other._indptr[:-1])
+        row = torch.cat([r1, r2])
+        col = torch.cat([self._indices.long(), other._indices.long()])
+        val = torch.cat([self._data.to(dt), alpha * other._data.to(dt)])
+        key = row * self._shape[1] + col
+        key, order = torch.sort(key, stable=True)
+        val = val[order]
+        uniq, inverse = torch.unique_consecutive(key, return_inverse=True)
+        out_val = torch.zeros(uniq.numel(), dtype=dt, device=dev)
+        out_val.index_add_(0, inverse, val)
+        out_row = torch.div(uniq, self._shape[1], rounding_mode="floor")
+        out_col = uniq - out_row * self._shape[1]
+        cnt = torch.bincount(out_row, minlength=lr)
+        indptr = torch.zeros(lr + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(cnt, dim=0, out=indptr[1:])
+        out = csr_array.__new__(csr_array)
+        out._init_local(indptr, out_col, out_val.contiguous(), self._shape)
+        return out
+
+    def __add__(self, other):
+        if isinstance(other, csr_array):
+            return self._add_sparse(other, 1.0)
+        if hasattr(other, "tocsr"):
+            return self._add_sparse(csr_array(other), 1.0)
+        if np.isscalar(other) and other == 0:
+            return self.copy()
+        return NotImplemented
+
+    __radd__ = __add__
+
+    def __sub__(self, other):
+        if isinstance(other, csr_array):
+            return self._add_sparse(other, -1.0)
+        if hasattr(other, "tocsr"):
+            return self._add_sparse(csr_array(other), -1.0)
+        return NotImplemented
+
+    def eliminate_zeros(self):
+        """Drop explicitly stored zeros in place (scipy API)."""
+        keep = self._data != 0
+        dev = self._data.device
+        lr = self._row_hi - self._row_lo
+        row = torch.repeat_interleave(
+            torch.arange(lr, device=dev),
+            self._indptr[1:] - self._indptr[:-1])[keep]
+        self._indices = self._indices[keep].contiguous()
+        self._data = self._data[keep].contiguous()
+        cnt = torch.bincount(row, minlength=lr)
+        indptr = torch.zeros(lr + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(cnt, dim=0, out=indptr[1:])
+        self._indptr = indptr
+        self._global_nnz = None
+        self._halo_cache = None
+        self._ov_cache = None
+        return self
+
+    def sum_duplicates(self):
+        """No-op: construction keeps canonical form (scipy API)."""
+        return self
+
     def diagonal(self, k: int = 0):
         """Main diagonal (k=0 only, reference csr.py:345-368); returns the
         LOCAL shard of the diagonal under partition(M)."""

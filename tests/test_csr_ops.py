@@ -126,3 +126,28 @@ def test_rmatmul_and_flags():
     np.testing.assert_allclose(to_np(x @ A), x @ S, rtol=1e-12)
     assert A.has_sorted_indices
     assert A.has_canonical_format
+
+
+def test_add_sub_sparse():
+    A = sample_csr(11, 9, 0.3, seed=21)
+    B = sample_csr(11, 9, 0.35, seed=22)
+    La, Lb = lsp.csr_array(A), lsp.csr_array(B)
+    np.testing.assert_allclose(to_np((La + Lb).todense()),
+                               (A + B).toarray(), rtol=1e-12, atol=1e-14)
+    np.testing.assert_allclose(to_np((La - Lb).todense()),
+                               (A - B).toarray(), rtol=1e-12, atol=1e-14)
+    with pytest.raises(ValueError):
+        La + lsp.csr_array((3, 3))
+
+
+def test_eliminate_zeros():
+    D = np.array([[1.0, 0, 2], [0, 0, 0], [3, 4, 0]])
+    A = lsp.csr_array(D)
+    A.data[1] = 0.0  # make one stored entry explicitly zero
+    A.eliminate_zeros()
+    S = sp.csr_array(D)
+    S.data[1] = 0.0
+    S.eliminate_zeros()
+    assert A.nnz == S.nnz
+    np.testing.assert_allclose(to_np(A.todense()), S.toarray())
+    A.sum_duplicates()  # no-op
