@@ -1,0 +1,12 @@
+#!/usr/bin/env python3
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+from legate_sparse.gallery import poisson_2d
+from legate_sparse import ops
+A = poisson_2d(4096, 4096)
+x = torch.rand(A.shape[0], dtype=torch.float64, device="cuda")
+y = torch.empty_like(x)
+for _ in range(30):
+    ops.spmv(A._indptr, A._indices, A._data, x, y)
+torch.cuda.synchronize()
