@@ -110,6 +110,74 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_pair_kernel(
   }
 }
 
+// PAIR2 variant: each W-lane group serves TWO ADJACENT rows per step,
+// interleaving their pair loads — 94.5% of the PAIR kernel's wave
+// cycles are parked on memory waits (SQ_WAIT_ANY), so doubling the
+// independent load chains per lane raises memory-level parallelism.
+template <typename T, typename I, int W, bool SWZ>
+__global__ __launch_bounds__(LS_THREADS) void spmv_pair2_kernel(
+    const i64* __restrict__ indptr, const I* __restrict__ indices,
+    const T* __restrict__ vals, const T* __restrict__ x, T* __restrict__ y,
+    i64 n_rows, int accumulate) {
+  using T2 = __attribute__((ext_vector_type(2))) T;
+  using I2 = __attribute__((ext_vector_type(2))) I;
+  constexpr int GROUPS = LS_THREADS / W;
+  const int group = threadIdx.x / W;
+  const int lane = threadIdx.x % W;
+  const int blk = SWZ ? xcd_swizzle(blockIdx.x, gridDim.x) : blockIdx.x;
+  const i64 stride = (i64)gridDim.x * GROUPS * 2;
+  for (i64 row = ((i64)blk * GROUPS + group) * 2; row < n_rows;
+       row += stride) {
+    const bool has2 = row + 1 < n_rows;
+    const i64 s1 = indptr[row];
+    const i64 e1 = indptr[row + 1];
+    const i64 s2 = has2 ? e1 : 0;
+    const i64 e2 = has2 ? indptr[row + 2] : 0;
+    const i64 b1 = s1 & ~(i64)1;
+    const i64 b2 = s2 & ~(i64)1;
+    T acc1 = ls_zero<T>();
+    T acc2 = ls_zero<T>();
+    i64 p1 = b1 + 2 * lane;
+    i64 p2 = b2 + 2 * lane;
+    while (p1 < e1 || p2 < e2) {
+      // issue both rows' loads before either use
+      T2 v1, v2;
+      I2 c1, c2;
+      const bool g1 = p1 < e1;
+      const bool g2 = p2 < e2;
+      if (g1) {
+        v1 = *reinterpret_cast<const T2*>(vals + p1);
+        c1 = *reinterpret_cast<const I2*>(indices + p1);
+      }
+      if (g2) {
+        v2 = *reinterpret_cast<const T2*>(vals + p2);
+        c2 = *reinterpret_cast<const I2*>(indices + p2);
+      }
+      if (g1) {
+        if (p1 >= s1) acc1 += v1.x * x[c1.x];
+        if (p1 + 1 < e1) acc1 += v1.y * x[c1.y];
+        p1 += 2 * W;
+      }
+      if (g2) {
+        if (p2 >= s2) acc2 += v2.x * x[c2.x];
+        if (p2 + 1 < e2) acc2 += v2.y * x[c2.y];
+        p2 += 2 * W;
+      }
+    }
+    acc1 = group_reduce_sum<T, W>(acc1);
+    acc2 = group_reduce_sum<T, W>(acc2);
+    if (lane == 0) {
+      if (accumulate) {
+        y[row] += acc1;
+        if (has2) y[row + 1] += acc2;
+      } else {
+        y[row] = acc1;
+        if (has2) y[row + 1] = acc2;
+      }
+    }
+  }
+}
+
 // STREAM variant (CSR-stream): each block owns a fixed ELEMENT range
 // [base, base+NB), loads val/idx pairs coalesced, gathers x, stages the
 // products in LDS, then reduces its rows from LDS (binary search for the
@@ -216,6 +284,15 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
     constexpr int RPB = LS_THREADS / WS;
     int grid = grid_1d(n_rows, RPB, gcap);
     if constexpr (!is_cplx<T>::value) {
+      if (pair_mode == 4) {
+        constexpr int RPB2 = (LS_THREADS / WS) * 2;
+        int grid2 = grid_1d(n_rows, RPB2, gcap);
+        hipLaunchKernelGGL((spmv_pair2_kernel<T, I, WS, false>),
+                           dim3(grid2), dim3(LS_THREADS), 0, stream,
+                           indptr, indices, vals, x, y, n_rows,
+                           accumulate ? 1 : 0);
+        return;
+      }
       if (pair) {
         if (swz)
           hipLaunchKernelGGL((spmv_pair_kernel<T, I, WS, true>),
