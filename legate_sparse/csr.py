@@ -206,15 +206,13 @@ class csr_array(CompressedBase):
         keep = (row >= lo) & (row < hi)
         row, col, data = row[keep] - lo, col[keep], data[keep]
         row, col, data = row.to(device), col.to(device), data.to(device)
-        # sort by (row, col): stable argsort over col then row
-        # (reference csr.py:204-214 uses stable argsort + bincount + cumsum)
-        order = torch.argsort(col, stable=True)
-        row, col, data = row[order], col[order], data[order]
-        order = torch.argsort(row, stable=True)
-        row, col, data = row[order], col[order], data[order]
-        # sum duplicates (scipy semantic)
+        # sort by (row, col) via one fused-key argsort (the reference does
+        # two stable argsorts, csr.py:204-214)
         if row.numel():
             key = row * N + col
+            order = torch.argsort(key)
+            row, col, data, key = (row[order], col[order], data[order],
+                                   key[order])
             uniq, inverse = torch.unique_consecutive(key, return_inverse=True)
             if uniq.numel() != key.numel():
                 newdata = torch.zeros(uniq.numel(), dtype=data.dtype,
@@ -650,13 +648,12 @@ class csr_array(CompressedBase):
             grow = torch.cat(comm.alltoallv(splits_r))
             col = torch.cat(comm.alltoallv(splits_c))
             val = torch.cat(comm.alltoallv(splits_v))
-        # local assembly: new rows = col - new_lo
+        # local assembly: new rows = col - new_lo; one fused-key argsort
+        # ((row, col) pairs are unique, so stability is not needed)
         new_lo = newpart.lo(runtime.rank)
         nrow = col - new_lo
         ncol = grow
-        order = torch.argsort(ncol, stable=True)
-        nrow, ncol, val = nrow[order], ncol[order], val[order]
-        order = torch.argsort(nrow, stable=True)
+        order = torch.argsort(nrow * M + ncol)
         nrow, ncol, val = nrow[order], ncol[order], val[order]
         lrn = newpart.count(runtime.rank)
         cnt = torch.bincount(nrow, minlength=lrn)
