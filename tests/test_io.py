@@ -1,6 +1,7 @@
 # SPDX-License-Identifier: Apache-2.0
 """mmread vs scipy.io.mmread (reference test_io.py semantics)."""
 import io
+import os
 
 import numpy as np
 import pytest
@@ -76,3 +77,17 @@ def test_save_load_npz(tmp_path):
     B = lsp2.load_npz(path)
     assert B.shape == A.shape
     np.testing.assert_allclose(to_np(B.todense()), S.toarray())
+
+
+REPO_TESTDATA = os.path.join(
+    os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    "testdata")
+
+
+@pytest.mark.parametrize("name", ["general_real.mtx", "symmetric.mtx",
+                                  "pattern.mtx", "complex_general.mtx"])
+def test_mmread_fixtures(name):
+    path = os.path.join(REPO_TESTDATA, name)
+    got = lsp.mmread(path)
+    want = scipy.io.mmread(path).tocsr()
+    np.testing.assert_allclose(to_np(got.todense()), want.toarray())
