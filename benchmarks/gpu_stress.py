@@ -22,12 +22,12 @@ from legate_sparse.settings import settings
 
 def rand_csr(rng, m, n, mode):
     if mode == "uniform":
-        density = rng.uniform(0.001, 0.2)
+        density = rng.uniform(0.001, 0.02)
         nnz = max(1, int(m * n * density))
         r = rng.integers(0, m, nnz)
         c = rng.integers(0, n, nnz)
     elif mode == "banded":
-        half = int(rng.integers(1, 30))
+        half = int(rng.integers(1, 12))
         offs = np.arange(-half, half + 1)
         r = np.repeat(np.arange(m), len(offs))
         c = r.reshape(m, -1) + offs
@@ -35,7 +35,7 @@ def rand_csr(rng, m, n, mode):
         r = np.repeat(np.arange(m), len(offs))[keep.ravel()]
         c = c.ravel()[keep.ravel()]
     else:  # powerlaw
-        nnz = max(1, int(m * rng.uniform(2, 30)))
+        nnz = max(1, int(m * rng.uniform(2, 12)))
         r = (m * rng.power(0.3, nnz)).astype(np.int64) % m
         c = (n * rng.power(0.3, nnz)).astype(np.int64) % n
     v = rng.standard_normal(len(r))
@@ -51,9 +51,9 @@ def main():
     rng = np.random.default_rng(args.seed)
 
     for rd in range(args.rounds):
-        m = int(rng.integers(16, 3000))
-        k = int(rng.integers(16, 3000))
-        n = int(rng.integers(16, 3000))
+        m = int(rng.integers(16, 900))
+        k = int(rng.integers(16, 900))
+        n = int(rng.integers(16, 900))
         mode = ["uniform", "banded", "powerlaw"][rd % 3]
         A = rand_csr(rng, m, k, mode)
         B = rand_csr(rng, k, n, mode)
