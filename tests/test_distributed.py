@@ -58,3 +58,26 @@ def test_gmg_distributed_world2():
     # exercises distributed SpGEMM (R A P), transpose, diagonal, CG+V-cycle
     out = _run_example_dist(2, "gmg.py", "-k", "5")
     assert "PASS" in out
+
+
+def _run_dist_env(nproc, env_extra, timeout=420):
+    env = dict(os.environ)
+    env.setdefault("MASTER_ADDR", "127.0.0.1")
+    env.update(env_extra)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    worker = os.path.join(repo, "tests", "dist_worker.py")
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           f"--nproc-per-node={nproc}", "--master-addr", "127.0.0.1",
+           "--master-port", "29775", worker]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=timeout,
+                       env=env, cwd=repo)
+    assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
+    assert "DIST_ALL_OK" in r.stdout
+
+
+def test_distributed_fast_spgemm():
+    _run_dist_env(2, {"LS_FAST_SPGEMM": "1"})
+
+
+def test_distributed_coarse_images():
+    _run_dist_env(2, {"LS_PRECISE_IMAGES": "0"})
