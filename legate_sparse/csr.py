@@ -71,6 +71,31 @@ def _common_dtype(*tensors) -> torch.dtype:
     return dt
 
 
+def _assemble_local_rows(row_local, col, val, lr, N, dedup=True):
+    """Sort (row, col), optionally sum duplicates, build local indptr.
+    Shared by the COO constructor, transpose, and gallery.rmat."""
+    dev = val.device
+    if row_local.numel():
+        key = row_local * N + col
+        order = torch.argsort(key)
+        row_local, col, val, key = (row_local[order], col[order],
+                                    val[order], key[order])
+        if dedup:
+            uniq, inverse = torch.unique_consecutive(key,
+                                                     return_inverse=True)
+            if uniq.numel() != key.numel():
+                newval = torch.zeros(uniq.numel(), dtype=val.dtype,
+                                     device=dev)
+                newval.index_add_(0, inverse, val)
+                val = newval
+                row_local = torch.div(uniq, N, rounding_mode="floor")
+                col = uniq - row_local * N
+    cnt = torch.bincount(row_local, minlength=lr)
+    indptr = torch.zeros(lr + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(cnt, dim=0, out=indptr[1:])
+    return indptr, col, val
+
+
 class csr_array(CompressedBase):
     """scipy.sparse.csr_array-compatible distributed CSR matrix."""
 
@@ -206,26 +231,11 @@ class csr_array(CompressedBase):
         keep = (row >= lo) & (row < hi)
         row, col, data = row[keep] - lo, col[keep], data[keep]
         row, col, data = row.to(device), col.to(device), data.to(device)
-        # sort by (row, col) via one fused-key argsort (the reference does
-        # two stable argsorts, csr.py:204-214)
-        if row.numel():
-            key = row * N + col
-            order = torch.argsort(key)
-            row, col, data, key = (row[order], col[order], data[order],
-                                   key[order])
-            uniq, inverse = torch.unique_consecutive(key, return_inverse=True)
-            if uniq.numel() != key.numel():
-                newdata = torch.zeros(uniq.numel(), dtype=data.dtype,
-                                      device=device)
-                newdata.index_add_(0, inverse, data)
-                data = newdata
-                row = torch.div(uniq, N, rounding_mode="floor")
-                col = uniq - row * N
         lr = hi - lo
-        counts = torch.bincount(row, minlength=lr)
-        indptr = torch.zeros(lr + 1, dtype=_INDEX_DTYPE, device=device)
-        torch.cumsum(counts, dim=0, out=indptr[1:])
-        self._init_local(indptr, col.to(_INDEX_DTYPE), data, (M, N))
+        indptr, col, data = _assemble_local_rows(row, col, data, lr, N,
+                                                 dedup=True)
+        self._init_local(indptr, col.to(_INDEX_DTYPE),
+                         data.contiguous(), (M, N))
 
     # ------------------------------------------------------------------
     # Properties (reference csr.py:289-336)
@@ -648,17 +658,11 @@ class csr_array(CompressedBase):
             grow = torch.cat(comm.alltoallv(splits_r))
             col = torch.cat(comm.alltoallv(splits_c))
             val = torch.cat(comm.alltoallv(splits_v))
-        # local assembly: new rows = col - new_lo; one fused-key argsort
-        # ((row, col) pairs are unique, so stability is not needed)
+        # local assembly: new rows = col - new_lo
         new_lo = newpart.lo(runtime.rank)
-        nrow = col - new_lo
-        ncol = grow
-        order = torch.argsort(nrow * M + ncol)
-        nrow, ncol, val = nrow[order], ncol[order], val[order]
         lrn = newpart.count(runtime.rank)
-        cnt = torch.bincount(nrow, minlength=lrn)
-        indptr = torch.zeros(lrn + 1, dtype=_INDEX_DTYPE, device=dev)
-        torch.cumsum(cnt, dim=0, out=indptr[1:])
+        indptr, ncol, val = _assemble_local_rows(
+            col - new_lo, grow, val, lrn, M, dedup=False)
         out = csr_array.__new__(csr_array)
         out._init_local(indptr, ncol.to(_INDEX_DTYPE), val.contiguous(),
                         (N, M))

@@ -161,22 +161,8 @@ def rmat(scale: int, edge_factor: int = 16, seed: int = 0,
         dst = torch.cat(comm.alltoallv(list(torch.split(dst, counts))))
         val = torch.cat(comm.alltoallv(list(torch.split(val, counts))))
     lo, hi = part.lo(rank), part.hi(rank)
-    row = src - lo
-    # sort by (row, col), then sum duplicates
-    order = torch.argsort(dst, stable=True)
-    row, dst, val = row[order], dst[order], val[order]
-    order = torch.argsort(row, stable=True)
-    row, dst, val = row[order], dst[order], val[order]
-    key = row * N + dst
-    uniq, inverse = torch.unique_consecutive(key, return_inverse=True)
-    if uniq.numel() != key.numel():
-        newval = torch.zeros(uniq.numel(), dtype=tdt, device=dev)
-        newval.index_add_(0, inverse, val)
-        val = newval
-        row = torch.div(uniq, N, rounding_mode="floor")
-        dst = uniq - row * N
-    lr = hi - lo
-    cnt = torch.bincount(row, minlength=lr)
-    indptr = torch.zeros(lr + 1, dtype=torch.int64, device=dev)
-    torch.cumsum(cnt, dim=0, out=indptr[1:])
-    return csr_array(None, shape=(N, N), _local=(indptr, dst, val))
+    from .csr import _assemble_local_rows
+    indptr, dst, val = _assemble_local_rows(src - lo, dst, val, hi - lo, N,
+                                            dedup=True)
+    return csr_array(None, shape=(N, N),
+                     _local=(indptr, dst, val.contiguous()))
