@@ -126,6 +126,11 @@ class dia_array(CompressedBase):
                         out_vals.contiguous(), (M, N))
         return out
 
+    def todense(self):
+        return self.tocsr().todense()
+
+    toarray = todense
+
     def __repr__(self):
         return (f"<legate_sparse.dia_array shape={self._shape} "
                 f"n_diags={self._offsets.numel()} dtype={self.dtype}>")
