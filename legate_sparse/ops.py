@@ -176,27 +176,26 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
                       bp_ptr, ub.data_ptr(), n_rows, icode, st)
     _tick("row_ub")
     cuts = ext.spgemm_lds_bins  # (48, 128, 1024, 4096)
-    # one-pass binning: MERGE bins (register k-way merge, rows with few
-    # A-entries and bounded expansion) then LDS-hash bins by expansion
-    # bound, then the HBM bin — one bucketize + argsort + one transfer
-    a_len = A_indptr[1:] - A_indptr[:-1]
-    medges = torch.tensor([8, 32, 64], dtype=torch.int64, device=dev)
-    mcut = torch.bucketize(a_len, medges, right=False)  # 0,1,2 / 3=no
-    mergeable = (mcut < 3) & (ub <= int(cuts[-1]))
-    edges = torch.tensor(list(cuts), dtype=torch.int64, device=dev)
-    bin_id = torch.where(mergeable, mcut,
-                         3 + torch.bucketize(ub, edges, right=False))
-    order = torch.argsort(bin_id.to(torch.int32), stable=True)
-    counts = torch.bincount(bin_id,
-                            minlength=len(cuts) + 4).cpu().tolist()
-    groups = []
-    off = 0
-    for c in counts[:len(cuts) + 3]:
-        groups.append(order[off:off + c].contiguous())
-        off += c
+    # one-pass binning in two tiny kernels (count + scatter; thresholds
+    # live in src/hip/spgemm.hip::row_bin): bins 0-2 merge W=8/32/64,
+    # 3-6 LDS-hash by expansion bound, 7 HBM.  Order within a bin is
+    # arbitrary — rows are independent.
+    counts_d = torch.zeros(8, dtype=torch.int64, device=dev)
+    ext.spgemm_bin_count(A_indptr.data_ptr(), ub.data_ptr(), n_rows,
+                         counts_d.data_ptr(), st)
+    counts = counts_d.cpu().tolist()
+    bases = [0]
+    for c in counts[:-1]:
+        bases.append(bases[-1] + int(c))
+    cursors = torch.tensor(bases, dtype=torch.int64, device=dev)
+    rows_out = torch.empty(n_rows, dtype=torch.int64, device=dev)
+    ext.spgemm_bin_scatter(A_indptr.data_ptr(), ub.data_ptr(), n_rows,
+                           cursors.data_ptr(), rows_out.data_ptr(), st)
+    groups = [rows_out[bases[i]:bases[i] + int(counts[i])]
+              for i in range(8)]
     mbins = groups[:3]          # merge W=8/32/64
-    bins = groups[3:]           # LDS-hash cfg0..3
-    rows_g = order[off:].contiguous()
+    bins = groups[3:7]          # LDS-hash cfg0..3
+    rows_g = groups[7].contiguous()
 
     row_nnz = torch.zeros(n_rows, dtype=torch.int64, device=dev)
     g_batches = []

@@ -15,6 +15,9 @@ void ls_spmv(uintptr_t, uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64,
              i64, int, int, bool, uintptr_t, int, bool, int, int);
 void ls_spgemm_row_ub(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, int,
                       uintptr_t);
+void ls_spgemm_bin_count(uintptr_t, uintptr_t, i64, uintptr_t, uintptr_t);
+void ls_spgemm_bin_scatter(uintptr_t, uintptr_t, i64, uintptr_t,
+                           uintptr_t, uintptr_t);
 void ls_spgemm_merge_symbolic(int, uintptr_t, i64, uintptr_t, uintptr_t,
                               uintptr_t, uintptr_t, uintptr_t, int,
                               uintptr_t);
@@ -64,6 +67,8 @@ PYBIND11_MODULE(_hip_kernels, m) {
   m.doc() = "legate_sparse gfx950 HIP kernels";
   m.def("spmv", &ls_spmv);
   m.def("spgemm_row_ub", &ls_spgemm_row_ub);
+  m.def("spgemm_bin_count", &ls_spgemm_bin_count);
+  m.def("spgemm_bin_scatter", &ls_spgemm_bin_scatter);
   m.def("spgemm_merge_symbolic", &ls_spgemm_merge_symbolic);
   m.def("spgemm_merge_numeric", &ls_spgemm_merge_numeric);
   m.def("spgemm_symbolic_lds", &ls_spgemm_symbolic_lds);

@@ -245,6 +245,60 @@ __global__ __launch_bounds__(BLOCK) void spgemm_numeric_lds_kernel(
 // vals pre-zeroed by the host).  Output UNSORTED; host post-sorts these
 // rare rows.
 // ---------------------------------------------------------------------------
+// One-pass row binning: bin id from (a_len, ub) thresholds, per-bin
+// counts, then scatter compaction — replaces a bucketize+argsort+
+// bincount chain of ~15 small torch launches + a host sync.
+// Bins: 0-2 merge (a_len <= 8/32/64 and ub <= 4096), 3-6 LDS hash by ub
+// (<=48/128/1024/4096), 7 global.  Order within a bin is arbitrary
+// (rows are independent).
+// ---------------------------------------------------------------------------
+constexpr int N_BINS = 8;
+
+__device__ inline int row_bin(i64 alen, i64 ubv) {
+  if (ubv <= 4096 && alen <= 64) {
+    if (alen <= 8) return 0;
+    if (alen <= 32) return 1;
+    return 2;
+  }
+  if (ubv <= 48) return 3;
+  if (ubv <= 128) return 4;
+  if (ubv <= 1024) return 5;
+  if (ubv <= 4096) return 6;
+  return 7;
+}
+
+__global__ __launch_bounds__(LS_THREADS) void spgemm_bin_count_kernel(
+    const i64* __restrict__ A_indptr, const i64* __restrict__ ub,
+    i64 n_rows, i64* __restrict__ counts) {
+  __shared__ i64 loc[N_BINS];
+  if (threadIdx.x < N_BINS) loc[threadIdx.x] = 0;
+  __syncthreads();
+  const i64 stride = (i64)gridDim.x * blockDim.x;
+  for (i64 r = (i64)blockIdx.x * blockDim.x + threadIdx.x; r < n_rows;
+       r += stride) {
+    const int b = row_bin(A_indptr[r + 1] - A_indptr[r], ub[r]);
+    atomicAdd(reinterpret_cast<unsigned long long*>(&loc[b]), 1ull);
+  }
+  __syncthreads();
+  if (threadIdx.x < N_BINS && loc[threadIdx.x])
+    atomicAdd(reinterpret_cast<unsigned long long*>(&counts[threadIdx.x]),
+              (unsigned long long)loc[threadIdx.x]);
+}
+
+__global__ __launch_bounds__(LS_THREADS) void spgemm_bin_scatter_kernel(
+    const i64* __restrict__ A_indptr, const i64* __restrict__ ub,
+    i64 n_rows, i64* __restrict__ cursors, i64* __restrict__ rows_out) {
+  const i64 stride = (i64)gridDim.x * blockDim.x;
+  for (i64 r = (i64)blockIdx.x * blockDim.x + threadIdx.x; r < n_rows;
+       r += stride) {
+    const int b = row_bin(A_indptr[r + 1] - A_indptr[r], ub[r]);
+    const i64 at = (i64)atomicAdd(
+        reinterpret_cast<unsigned long long*>(&cursors[b]), 1ull);
+    rows_out[at] = r;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Register k-way MERGE path for rows with few A-entries (a_len <= W) and
 // bounded expansion: lane l owns A-entry l and walks its (sorted) B row;
 // each output element is a wave-min over the lanes' cursors plus a masked
@@ -861,6 +915,32 @@ void merge_numeric_dispatch(int wcfg, const i64* rl, i64 n, const i64* ap,
   else L(std::integral_constant<int, 64>{});
 }
 }  // namespace
+
+void ls_spgemm_bin_count(uintptr_t A_indptr, uintptr_t ub, i64 n_rows,
+                         uintptr_t counts, uintptr_t stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  hipLaunchKernelGGL(spgemm_bin_count_kernel,
+                     dim3(grid_1d(n_rows, LS_THREADS, 4096)),
+                     dim3(LS_THREADS), 0, s,
+                     reinterpret_cast<const i64*>(A_indptr),
+                     reinterpret_cast<const i64*>(ub), n_rows,
+                     reinterpret_cast<i64*>(counts));
+  ls_check(hipGetLastError(), "spgemm_bin_count");
+}
+
+void ls_spgemm_bin_scatter(uintptr_t A_indptr, uintptr_t ub, i64 n_rows,
+                           uintptr_t cursors, uintptr_t rows_out,
+                           uintptr_t stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  hipLaunchKernelGGL(spgemm_bin_scatter_kernel,
+                     dim3(grid_1d(n_rows, LS_THREADS, 4096)),
+                     dim3(LS_THREADS), 0, s,
+                     reinterpret_cast<const i64*>(A_indptr),
+                     reinterpret_cast<const i64*>(ub), n_rows,
+                     reinterpret_cast<i64*>(cursors),
+                     reinterpret_cast<i64*>(rows_out));
+  ls_check(hipGetLastError(), "spgemm_bin_scatter");
+}
 
 void ls_spgemm_merge_symbolic(int wcfg, uintptr_t rows_list, i64 n,
                               uintptr_t A_indptr, uintptr_t A_indices,
