@@ -288,13 +288,33 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_bin_count_kernel(
 __global__ __launch_bounds__(LS_THREADS) void spgemm_bin_scatter_kernel(
     const i64* __restrict__ A_indptr, const i64* __restrict__ ub,
     i64 n_rows, i64* __restrict__ cursors, i64* __restrict__ rows_out) {
-  const i64 stride = (i64)gridDim.x * blockDim.x;
-  for (i64 r = (i64)blockIdx.x * blockDim.x + threadIdx.x; r < n_rows;
-       r += stride) {
+  // two-pass per-block aggregation: ONE global atomicAdd per bin per
+  // block (a naive per-row atomic on 8 shared cursors serializes at
+  // ~88 ops/us — 0.2 s for 16.7M rows)
+  __shared__ i64 cnt[N_BINS];
+  __shared__ i64 base[N_BINS];
+  const i64 chunk = (n_rows + gridDim.x - 1) / gridDim.x;
+  const i64 r0 = (i64)blockIdx.x * chunk;
+  const i64 r1 = min(r0 + chunk, n_rows);
+  if (threadIdx.x < N_BINS) cnt[threadIdx.x] = 0;
+  __syncthreads();
+  for (i64 r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
     const int b = row_bin(A_indptr[r + 1] - A_indptr[r], ub[r]);
-    const i64 at = (i64)atomicAdd(
-        reinterpret_cast<unsigned long long*>(&cursors[b]), 1ull);
-    rows_out[at] = r;
+    atomicAdd(reinterpret_cast<unsigned long long*>(&cnt[b]), 1ull);
+  }
+  __syncthreads();
+  if (threadIdx.x < N_BINS) {
+    base[threadIdx.x] = (i64)atomicAdd(
+        reinterpret_cast<unsigned long long*>(&cursors[threadIdx.x]),
+        (unsigned long long)cnt[threadIdx.x]);
+    cnt[threadIdx.x] = 0;
+  }
+  __syncthreads();
+  for (i64 r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+    const int b = row_bin(A_indptr[r + 1] - A_indptr[r], ub[r]);
+    const i64 pos = (i64)atomicAdd(
+        reinterpret_cast<unsigned long long*>(&cnt[b]), 1ull);
+    rows_out[base[b] + pos] = r;
   }
 }
 
@@ -933,7 +953,7 @@ void ls_spgemm_bin_scatter(uintptr_t A_indptr, uintptr_t ub, i64 n_rows,
                            uintptr_t stream) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   hipLaunchKernelGGL(spgemm_bin_scatter_kernel,
-                     dim3(grid_1d(n_rows, LS_THREADS, 4096)),
+                     dim3(grid_1d(n_rows, LS_THREADS, 1024)),
                      dim3(LS_THREADS), 0, s,
                      reinterpret_cast<const i64*>(A_indptr),
                      reinterpret_cast<const i64*>(ub), n_rows,
