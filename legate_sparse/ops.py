@@ -367,19 +367,12 @@ def _spgemm_hip(A_indptr, A_indices, A_vals, B_indptr, B_indices, B_vals,
         nnz_ptr = row_nnz.data_ptr()
 
     if hybrid:
-        # compact merged rows from the capacity layout into C
-        cnts_m = row_nnz[mrows]
-        tot_m = int(cnts_m.sum().item())
-        if tot_m:
-            src0 = cap_off[mrows]
-            dst0 = C_indptr[mrows]
-            seg = torch.cumsum(cnts_m, 0) - cnts_m
-            pos = (torch.arange(tot_m, device=dev)
-                   - torch.repeat_interleave(seg, cnts_m))
-            src = torch.repeat_interleave(src0, cnts_m) + pos
-            dst = torch.repeat_interleave(dst0, cnts_m) + pos
-            C_indices[dst] = Ci_cap[src]
-            C_vals[dst] = Cv_cap[src]
+        # compact merged rows from the capacity layout into C (kernel:
+        # the torch index-assembly version of this copy cost ~16 ms)
+        ext.spgemm_compact_rows(
+            mrows.data_ptr(), mrows.numel(), cap_off.data_ptr(),
+            C_indptr.data_ptr(), Ci_cap.data_ptr(), Cv_cap.data_ptr(),
+            C_indices.data_ptr(), C_vals.data_ptr(), code, icode, st)
         del Ci_cap, Cv_cap
     else:
         for wcfg, rows in enumerate(mbins):

@@ -318,6 +318,32 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_bin_scatter_kernel(
   }
 }
 
+// Copy per-row segments from a capacity layout into the final CSR
+// arrays (hybrid merge-fast compaction): one W-lane group per row,
+// coalesced, 2x row bytes of traffic.
+template <typename T, typename I>
+__global__ __launch_bounds__(LS_THREADS) void spgemm_compact_rows_kernel(
+    const i64* __restrict__ rows_list, i64 n_bin_rows,
+    const i64* __restrict__ src_off, const i64* __restrict__ dst_indptr,
+    const I* __restrict__ src_idx, const T* __restrict__ src_val,
+    I* __restrict__ dst_idx, T* __restrict__ dst_val) {
+  constexpr int W = 8;
+  constexpr int RPB = LS_THREADS / W;
+  const int group = threadIdx.x / W;
+  const int lane = threadIdx.x % W;
+  const i64 stride = (i64)gridDim.x * RPB;
+  for (i64 i = (i64)blockIdx.x * RPB + group; i < n_bin_rows; i += stride) {
+    const i64 row = rows_list[i];
+    const i64 s = src_off[row];
+    const i64 d = dst_indptr[row];
+    const i64 m = dst_indptr[row + 1] - d;
+    for (i64 t = lane; t < m; t += W) {
+      dst_idx[d + t] = src_idx[s + t];
+      dst_val[d + t] = src_val[s + t];
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Register k-way MERGE path for rows with few A-entries (a_len <= W) and
 // bounded expansion: lane l owns A-entry l and walks its (sorted) B row;
@@ -935,6 +961,25 @@ void merge_numeric_dispatch(int wcfg, const i64* rl, i64 n, const i64* ap,
   else L(std::integral_constant<int, 64>{});
 }
 }  // namespace
+
+void ls_spgemm_compact_rows(uintptr_t rows_list, i64 n,
+                            uintptr_t src_off, uintptr_t dst_indptr,
+                            uintptr_t src_idx, uintptr_t src_val,
+                            uintptr_t dst_idx, uintptr_t dst_val,
+                            int dtype, int idx_dtype, uintptr_t stream) {
+  hipStream_t s = reinterpret_cast<hipStream_t>(stream);
+  DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, hipLaunchKernelGGL(
+      (spgemm_compact_rows_kernel<val_t, idx_t>),
+      dim3(grid_1d(n, LS_THREADS / 8, 8192)), dim3(LS_THREADS), 0, s,
+      reinterpret_cast<const i64*>(rows_list), n,
+      reinterpret_cast<const i64*>(src_off),
+      reinterpret_cast<const i64*>(dst_indptr),
+      reinterpret_cast<const idx_t*>(src_idx),
+      reinterpret_cast<const val_t*>(src_val),
+      reinterpret_cast<idx_t*>(dst_idx),
+      reinterpret_cast<val_t*>(dst_val))));
+  ls_check(hipGetLastError(), "spgemm_compact_rows");
+}
 
 void ls_spgemm_bin_count(uintptr_t A_indptr, uintptr_t ub, i64 n_rows,
                          uintptr_t counts, uintptr_t stream) {
