@@ -70,8 +70,10 @@ def build_hip(force: bool = False) -> str:
     if not force and _newer(out, srcs):
         return out
     hipcc = os.environ.get("HIPCC", "hipcc")
+    debug = os.environ.get("LS_DEBUG") == "1"
+    flags = (["-O1", "-g", "-DLS_BOUNDS_CHECK"] if debug else ["-O3"])
     cmd = (
-        [hipcc, "--offload-arch=gfx950", "-O3", "-std=c++17", "-shared",
+        [hipcc, "--offload-arch=gfx950", *flags, "-std=c++17", "-shared",
          "-fPIC", "-fvisibility=hidden", "-Wno-unused-result"]
         + _includes()
         + ["-x", "hip"]

@@ -16,6 +16,17 @@
 #define WAVE_SIZE 64
 #define LS_THREADS 256
 
+// Debug bounds checking (build with LS_DEBUG=1 -> -DLS_BOUNDS_CHECK):
+// aborts the kernel with a device-side trap on an out-of-range index —
+// the analogue of the reference's Legion_BOUNDS_CHECKS accessor checks
+// (SURVEY §5).  Pair with AMD_SERIALIZE_KERNEL=3 to attribute the trap
+// to a launch site.
+#ifdef LS_BOUNDS_CHECK
+#define LS_ASSERT_RANGE(i, n)                                     do {                                                              if ((i64)(i) < 0 || (i64)(i) >= (i64)(n)) {                       __builtin_trap();                                             }                                                             } while (0)
+#else
+#define LS_ASSERT_RANGE(i, n)   do {                          } while (0)
+#endif
+
 using i64 = int64_t;
 
 static inline void ls_check(hipError_t e, const char* what) {

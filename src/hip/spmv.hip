@@ -58,6 +58,7 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_vector_kernel(
     const i64 e = indptr[row + 1];
     T acc = ls_zero<T>();
     for (i64 jp = s + lane; jp < e; jp += W) {
+      LS_ASSERT_RANGE(indices[jp], (i64)1 << 62);
       if constexpr (NT)
         acc += nt_load(vals + jp) * x[nt_load(indices + jp)];
       else
