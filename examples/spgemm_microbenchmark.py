@@ -23,6 +23,10 @@ def main():
     ap.add_argument("--nnz-per-row", type=int, default=5)
     ap.add_argument("-i", "--iters", type=int, default=10)
     ap.add_argument("-w", "--warmup", type=int, default=2)
+    ap.add_argument("--fresh", action="store_true",
+                    help="rebuild the operands every iteration (the "
+                         "reference's non --stable mode: includes "
+                         "construction and plan setup)")
     args = ap.parse_args()
 
     A = banded_matrix(args.n, args.nnz_per_row, seed=1)
@@ -32,7 +36,10 @@ def main():
     c_nnz = C.nnz
     del C
     with Timer() as t:
-        for _ in range(args.iters):
+        for it in range(args.iters):
+            if args.fresh:
+                A = banded_matrix(args.n, args.nnz_per_row, seed=1)
+                B = banded_matrix(args.n, args.nnz_per_row, seed=2)
             C = A @ B
             del C
     per = t.elapsed / args.iters
