@@ -111,6 +111,46 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_pair_kernel(
   }
 }
 
+// STENCIL variant for uniformly short rows (max nnz/row <= 2*NP-1):
+// one lane per row, ALL pair loads issued unrolled before any use —
+// maximum per-lane memory-level parallelism, no sub-wave reduce.
+template <typename T, typename I, int NP>
+__global__ __launch_bounds__(LS_THREADS) void spmv_sten_kernel(
+    const i64* __restrict__ indptr, const I* __restrict__ indices,
+    const T* __restrict__ vals, const T* __restrict__ x, T* __restrict__ y,
+    i64 n_rows, int accumulate) {
+  using T2 = __attribute__((ext_vector_type(2))) T;
+  using I2 = __attribute__((ext_vector_type(2))) I;
+  const i64 stride = (i64)gridDim.x * blockDim.x;
+  for (i64 row = (i64)blockIdx.x * blockDim.x + threadIdx.x; row < n_rows;
+       row += stride) {
+    const i64 s = indptr[row];
+    const i64 e = indptr[row + 1];
+    const i64 s2 = s & ~(i64)1;
+    T2 v[NP];
+    I2 c[NP];
+#pragma unroll
+    for (int p = 0; p < NP; ++p) {
+      const i64 pp = s2 + 2 * p;
+      if (pp < e) {
+        v[p] = *reinterpret_cast<const T2*>(vals + pp);
+        c[p] = *reinterpret_cast<const I2*>(indices + pp);
+      }
+    }
+    T acc = ls_zero<T>();
+#pragma unroll
+    for (int p = 0; p < NP; ++p) {
+      const i64 pp = s2 + 2 * p;
+      if (pp >= s && pp < e) acc += v[p].x * x[c[p].x];
+      if (pp + 1 >= s && pp + 1 < e) acc += v[p].y * x[c[p].y];
+    }
+    if (accumulate)
+      y[row] += acc;
+    else
+      y[row] = acc;
+  }
+}
+
 // PAIR2 variant: each W-lane group serves TWO ADJACENT rows per step,
 // interleaving their pair loads — 94.5% of the PAIR kernel's wave
 // cycles are parked on memory waits (SQ_WAIT_ANY), so doubling the
@@ -285,6 +325,18 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
     constexpr int RPB = LS_THREADS / WS;
     int grid = grid_1d(n_rows, RPB, gcap);
     if constexpr (!is_cplx<T>::value) {
+      if (pair_mode == 5 || pair_mode == 6) {
+        int grid3 = grid_1d(n_rows, LS_THREADS, gcap);
+        if (pair_mode == 5)
+          hipLaunchKernelGGL((spmv_sten_kernel<T, I, 4>), dim3(grid3),
+                             dim3(LS_THREADS), 0, stream, indptr, indices,
+                             vals, x, y, n_rows, accumulate ? 1 : 0);
+        else
+          hipLaunchKernelGGL((spmv_sten_kernel<T, I, 8>), dim3(grid3),
+                             dim3(LS_THREADS), 0, stream, indptr, indices,
+                             vals, x, y, n_rows, accumulate ? 1 : 0);
+        return;
+      }
       if (pair_mode == 4) {
         constexpr int RPB2 = (LS_THREADS / WS) * 2;
         int grid2 = grid_1d(n_rows, RPB2, gcap);
