@@ -324,6 +324,16 @@ class csr_array(CompressedBase):
     # ------------------------------------------------------------------
     # Core ops
     # ------------------------------------------------------------------
+    def _max_row_nnz(self) -> int:
+        """Cached max row length (drives the stencil SpMV variant)."""
+        if getattr(self, "_max_nnz_cache", None) is None:
+            if self._indptr.numel() > 1:
+                self._max_nnz_cache = int(
+                    (self._indptr[1:] - self._indptr[:-1]).max())
+            else:
+                self._max_nnz_cache = 0
+        return self._max_nnz_cache
+
     def _halo_plan(self):
         """MIN_MAX column-window halo plan (the explicit replacement of
         the reference's image(crd, x, MIN_MAX) constraint, csr.py:591).
@@ -404,8 +414,9 @@ class csr_array(CompressedBase):
             cnt = torch.bincount(row_ids[m], minlength=lr)
             ip = torch.zeros(lr + 1, dtype=torch.int64, device=dev)
             torch.cumsum(cnt, dim=0, out=ip[1:])
-            return ip, self._indices[m].contiguous(), \
-                self._data[m].contiguous()
+            mx = int(cnt.max()) if cnt.numel() else 0
+            return (ip, self._indices[m].contiguous(),
+                    self._data[m].contiguous(), mx)
 
         self._ov_cache = (build(mask), build(~mask))
         return self._ov_cache
@@ -428,19 +439,21 @@ class csr_array(CompressedBase):
         N = self._shape[1]
         part = runtime.partition(N)
         lo = part.lo(runtime.rank)
-        (ip_i, ix_i, dv_i), (ip_h, ix_h, dv_h) = self._split_for_overlap()
+        ((ip_i, ix_i, dv_i, mx_i),
+         (ip_h, ix_h, dv_h, mx_h)) = self._split_for_overlap()
         if ix_h.numel() == 0:
-            return ops.spmv(ip_i, ix_i, dv_i, x, y, col_offset=lo)
+            return ops.spmv(ip_i, ix_i, dv_i, x, y, col_offset=lo,
+                            max_nnz=mx_i)
         plan = self._halo_plan()
         pieces = [x[s:s + c] for (s, c) in plan["send"] if c > 0]
         send = torch.cat(pieces) if pieces else x[:0]
         x_win, work = comm.alltoallv_single_async(
             send, [c for (_, c) in plan["send"]], plan["recv_counts"])
-        y = ops.spmv(ip_i, ix_i, dv_i, x, y, col_offset=lo)
+        y = ops.spmv(ip_i, ix_i, dv_i, x, y, col_offset=lo, max_nnz=mx_i)
         if work is not None:
             work.wait()
         ops.spmv(ip_h, ix_h, dv_h, x_win, y, accumulate=True,
-                 col_offset=plan["min_col"])
+                 col_offset=plan["min_col"], max_nnz=mx_h)
         return y
 
     def _coerce_vec(self, other) -> torch.Tensor:
@@ -486,7 +499,7 @@ class csr_array(CompressedBase):
         else:
             x_win, col_off = A._gather_x(x)
             y = ops.spmv(A._indptr, A._indices, A._data, x_win, y,
-                         col_offset=col_off)
+                         col_offset=col_off, max_nnz=A._max_row_nnz())
         return y.reshape(-1, 1) if squeezed else y
 
     def __matmul__(self, other):
@@ -730,7 +743,7 @@ def spmv(A: csr_array, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         return A._matvec_dist(xv, y.reshape(-1))
     x_win, col_off = A._gather_x(xv)
     return ops.spmv(A._indptr, A._indices, A._data, x_win, y.reshape(-1),
-                    col_offset=col_off)
+                    col_offset=col_off, max_nnz=A._max_row_nnz())
 
 
 def _gather_B_window(A: csr_array, B: csr_array):

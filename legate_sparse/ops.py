@@ -52,7 +52,7 @@ def spmv(indptr: torch.Tensor, indices: torch.Tensor, vals: torch.Tensor,
          x: torch.Tensor, y: Optional[torch.Tensor] = None,
          accumulate: bool = False, w_override: int = 0,
          nt: bool = False, pair: int = -1, swz: int = 0,
-         col_offset: int = 0) -> torch.Tensor:
+         col_offset: int = 0, max_nnz: int = -1) -> torch.Tensor:
     """col_offset: kernels read x[c - col_offset] for global column c —
     realized as an adjusted base pointer, the same "fake offset dense
     pointer" trick the reference plays on cuSPARSE (spmv.cu:75-90)."""
@@ -66,7 +66,8 @@ def spmv(indptr: torch.Tensor, indices: torch.Tensor, vals: torch.Tensor,
         ext.spmv(indptr.data_ptr(), indices.data_ptr(), vals.data_ptr(),
                  x_ptr, y.data_ptr(), n_rows, vals.numel(),
                  _code(vals), _icode(indices), accumulate, _stream(),
-                 int(w_override), bool(nt), int(pair), int(swz))
+                 int(w_override), bool(nt), int(pair), int(swz),
+                 int(max_nnz))
         return y
     if not vals.is_cuda and _cext.has_cpu():
         _cext.require_cpu().spmv(indptr.data_ptr(), indices.data_ptr(),

@@ -12,7 +12,7 @@ namespace py = pybind11;
 using i64 = int64_t;
 
 void ls_spmv(uintptr_t, uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64,
-             i64, int, int, bool, uintptr_t, int, bool, int, int);
+             i64, int, int, bool, uintptr_t, int, bool, int, int, i64);
 void ls_spgemm_row_ub(uintptr_t, uintptr_t, uintptr_t, uintptr_t, i64, int,
                       uintptr_t);
 void ls_spgemm_bin_count(uintptr_t, uintptr_t, i64, uintptr_t, uintptr_t);

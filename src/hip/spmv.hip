@@ -284,7 +284,7 @@ template <typename T, typename I>
 void spmv_launch(const i64* indptr, const I* indices, const T* vals,
                  const T* x, T* y, i64 n_rows, i64 nnz, bool accumulate,
                  int w_override, bool nt, int pair_mode, int swz_mode,
-                 hipStream_t stream) {
+                 i64 max_nnz, hipStream_t stream) {
   const double mean = n_rows > 0 ? (double)nnz / (double)n_rows : 0.0;
   // W = largest power of two <= mean/2 (measured: ~2 elements per lane,
   // profiles/spmv_sweep_r01.txt)
@@ -292,6 +292,11 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
   while (W < 64 && (double)(W * 4) <= mean) W *= 2;
   if (w_override > 0) W = w_override;
   constexpr bool is_real = !is_cplx<T>::value;
+  // uniformly short rows: the fully unrolled stencil kernel measured
+  // 592 vs 525 GF/s on 5-pt Poisson (profiles/spmv_pmc_r01.md)
+  if (is_real && pair_mode < 0 && max_nnz >= 0 && max_nnz <= 7) {
+    pair_mode = 5;
+  }
   const bool pair = (pair_mode < 0) ? (is_real && mean >= 3.0)
                                     : (pair_mode == 1 && is_real);
   const bool swz = swz_mode == 1;
@@ -390,7 +395,8 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
 void ls_spmv(uintptr_t indptr, uintptr_t indices, uintptr_t vals,
              uintptr_t x, uintptr_t y, i64 n_rows, i64 nnz, int dtype,
              int idx_dtype, bool accumulate, uintptr_t stream,
-             int w_override, bool nt, int pair_mode, int swz_mode) {
+             int w_override, bool nt, int pair_mode, int swz_mode,
+             i64 max_nnz) {
   hipStream_t s = reinterpret_cast<hipStream_t>(stream);
   DISPATCH_VAL_T(dtype, DISPATCH_IDX_T(idx_dtype, (
       spmv_launch<val_t, idx_t>(
@@ -399,5 +405,5 @@ void ls_spmv(uintptr_t indptr, uintptr_t indices, uintptr_t vals,
           reinterpret_cast<const val_t*>(vals),
           reinterpret_cast<const val_t*>(x), reinterpret_cast<val_t*>(y),
           n_rows, nnz, accumulate, w_override, nt, pair_mode, swz_mode,
-          s))));
+          max_nnz, s))));
 }
