@@ -151,6 +151,52 @@ __global__ __launch_bounds__(LS_THREADS) void spmv_sten_kernel(
   }
 }
 
+// PAIRU variant: W lanes per row, each lane's PPL pairs fully unrolled
+// (coverage: max nnz + 1 <= 2*W*PPL).  Generalizes the stencil kernel
+// to wider rows while keeping every load issued before any use.
+template <typename T, typename I, int W, int PPL>
+__global__ __launch_bounds__(LS_THREADS) void spmv_pairu_kernel(
+    const i64* __restrict__ indptr, const I* __restrict__ indices,
+    const T* __restrict__ vals, const T* __restrict__ x, T* __restrict__ y,
+    i64 n_rows, int accumulate) {
+  using T2 = __attribute__((ext_vector_type(2))) T;
+  using I2 = __attribute__((ext_vector_type(2))) I;
+  constexpr int RPB = LS_THREADS / W;
+  const int group = threadIdx.x / W;
+  const int lane = threadIdx.x % W;
+  const i64 stride = (i64)gridDim.x * RPB;
+  for (i64 row = (i64)blockIdx.x * RPB + group; row < n_rows;
+       row += stride) {
+    const i64 s = indptr[row];
+    const i64 e = indptr[row + 1];
+    const i64 s2 = s & ~(i64)1;
+    T2 v[PPL];
+    I2 c[PPL];
+#pragma unroll
+    for (int k = 0; k < PPL; ++k) {
+      const i64 pp = s2 + 2 * (lane + k * W);
+      if (pp < e) {
+        v[k] = *reinterpret_cast<const T2*>(vals + pp);
+        c[k] = *reinterpret_cast<const I2*>(indices + pp);
+      }
+    }
+    T acc = ls_zero<T>();
+#pragma unroll
+    for (int k = 0; k < PPL; ++k) {
+      const i64 pp = s2 + 2 * (lane + k * W);
+      if (pp >= s && pp < e) acc += v[k].x * x[c[k].x];
+      if (pp + 1 >= s && pp + 1 < e) acc += v[k].y * x[c[k].y];
+    }
+    acc = group_reduce_sum<T, W>(acc);
+    if (lane == 0) {
+      if (accumulate)
+        y[row] += acc;
+      else
+        y[row] = acc;
+    }
+  }
+}
+
 // PAIR2 variant: each W-lane group serves TWO ADJACENT rows per step,
 // interleaving their pair loads — 94.5% of the PAIR kernel's wave
 // cycles are parked on memory waits (SQ_WAIT_ANY), so doubling the
@@ -292,10 +338,16 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
   while (W < 64 && (double)(W * 4) <= mean) W *= 2;
   if (w_override > 0) W = w_override;
   constexpr bool is_real = !is_cplx<T>::value;
-  // uniformly short rows: the fully unrolled stencil kernel measured
-  // 592 vs 525 GF/s on 5-pt Poisson (profiles/spmv_pmc_r01.md)
-  if (is_real && pair_mode < 0 && max_nnz >= 0 && max_nnz <= 7) {
-    pair_mode = 5;
+  // uniformly short rows: fully unrolled kernels (every load issued
+  // before any use) measured 592 vs 525 GF/s on 5-pt Poisson
+  // (profiles/spmv_pmc_r01.md)
+  if (is_real && pair_mode < 0 && max_nnz >= 0) {
+    if (max_nnz <= 7)
+      pair_mode = 5;        // stencil W=1, PPL=4
+    else if (max_nnz <= 15)
+      pair_mode = 7;        // pairu W=2, PPL=4
+    else if (max_nnz <= 31)
+      pair_mode = 8;        // pairu W=4, PPL=4
   }
   const bool pair = (pair_mode < 0) ? (is_real && mean >= 3.0)
                                     : (pair_mode == 1 && is_real);
@@ -330,6 +382,22 @@ void spmv_launch(const i64* indptr, const I* indices, const T* vals,
     constexpr int RPB = LS_THREADS / WS;
     int grid = grid_1d(n_rows, RPB, gcap);
     if constexpr (!is_cplx<T>::value) {
+      if (pair_mode == 7) {
+        constexpr int RPB7 = LS_THREADS / 2;
+        hipLaunchKernelGGL((spmv_pairu_kernel<T, I, 2, 4>),
+                           dim3(grid_1d(n_rows, RPB7, gcap)),
+                           dim3(LS_THREADS), 0, stream, indptr, indices,
+                           vals, x, y, n_rows, accumulate ? 1 : 0);
+        return;
+      }
+      if (pair_mode == 8) {
+        constexpr int RPB8 = LS_THREADS / 4;
+        hipLaunchKernelGGL((spmv_pairu_kernel<T, I, 4, 4>),
+                           dim3(grid_1d(n_rows, RPB8, gcap)),
+                           dim3(LS_THREADS), 0, stream, indptr, indices,
+                           vals, x, y, n_rows, accumulate ? 1 : 0);
+        return;
+      }
       if (pair_mode == 5 || pair_mode == 6) {
         int grid3 = grid_1d(n_rows, LS_THREADS, gcap);
         if (pair_mode == 5)
