@@ -41,7 +41,8 @@ def test_gpu_spmv_dtypes(dtype):
     np.testing.assert_allclose(to_np(y), S @ x, rtol=tol, atol=tol)
 
 
-@pytest.mark.parametrize("nnz_per_row", [1, 3, 5, 11, 33, 101])
+@pytest.mark.parametrize("nnz_per_row",
+                         [1, 3, 5, 7, 9, 11, 15, 17, 31, 33, 101])
 def test_gpu_spmv_row_lengths(nnz_per_row):
     # sweeps the W (sub-wave width) selection paths of spmv.hip
     n = 2000
