@@ -429,15 +429,9 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_merge_numeric_kernel(
     kp = B_indptr[k];
     kend = B_indptr[k + 1];
   }
-  // one-deep lookahead: the element consumed by grp_min is always
-  // already in registers; the advance issues the NEXT load, whose
-  // latency hides under the min/sum shuffle trees of this iteration
   uint32_t mycol = (lane < alen && kp < kend) ? (uint32_t)B_indices[kp]
                                               : EMPTY;
   T myval = (mycol != EMPTY) ? B_vals[kp] : ls_zero<T>();
-  uint32_t ncol = (mycol != EMPTY && kp + 1 < kend)
-                      ? (uint32_t)B_indices[kp + 1] : EMPTY;
-  T nval = (ncol != EMPTY) ? B_vals[kp + 1] : ls_zero<T>();
   const i64 base = C_indptr[row];
   i64 out = 0;
   while (true) {
@@ -452,13 +446,11 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_merge_numeric_kernel(
     ++out;
     if (mycol == mn) {
       ++kp;
-      mycol = ncol;
-      myval = nval;
-      if (kp + 1 < kend) {
-        ncol = (uint32_t)B_indices[kp + 1];
-        nval = B_vals[kp + 1];
+      if (kp < kend) {
+        mycol = (uint32_t)B_indices[kp];
+        myval = B_vals[kp];
       } else {
-        ncol = EMPTY;
+        mycol = EMPTY;
       }
     }
   }
