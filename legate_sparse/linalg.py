@@ -275,8 +275,7 @@ def cg(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
             torch.cuda.current_stream().wait_stream(side)
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
-                _one_iter()
-            iters += 1  # the capture run also executes
+                _one_iter()  # recorded, NOT executed (stream capture)
         except Exception:
             graph = None  # fall back to the eager loop
 
