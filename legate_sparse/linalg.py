@@ -139,6 +139,11 @@ def aslinearoperator(A):
     raise TypeError(f"cannot wrap {type(A)} as LinearOperator")
 
 
+# reference linalg.py:417-431 exposes make_linear_operator; scipy calls
+# the same thing aslinearoperator — provide both names.
+make_linear_operator = aslinearoperator
+
+
 def norm(A, ord="fro"):
     """Matrix norm of a sparse array: 'fro', 1 (max column abs sum), or
     inf (max row abs sum) — scipy.sparse.linalg.norm-compatible subset."""

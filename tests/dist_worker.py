@@ -127,6 +127,33 @@ def main():
     check("rmat spgemm", np.allclose(t2.numpy(), dense @ dense,
                                      rtol=1e-10, atol=1e-10))
 
+    # --- degenerate shapes: fewer rows than ranks (empty local shards),
+    # zero-nnz matrices, rows with no entries ---
+    n2 = max(2, ws - 1)  # at least one rank owns zero rows
+    S2 = sp.csr_array(np.triu(np.ones((n2, n2))))
+    A2 = lsp.csr_array(S2)
+    x2 = np.arange(1.0, n2 + 1)
+    y2f = to_np(lsu.full_vector(A2 @ x2, n2))
+    check("tiny spmv (empty shards)", np.allclose(y2f, S2 @ x2))
+    C2 = A2 @ A2
+    got2 = np.zeros((n2, n2))
+    lo2, hi2 = C2.row_range
+    if hi2 > lo2:
+        got2[lo2:hi2] = to_np(C2.todense())
+    t2 = torch.from_numpy(got2)
+    dist.all_reduce(t2)
+    check("tiny spgemm (empty shards)",
+          np.allclose(t2.numpy(), (S2 @ S2).toarray()))
+    check("tiny transpose", A2.T.nnz == S2.nnz)
+
+    Z = lsp.csr_array(sp.csr_array((np.zeros(0), np.zeros(0, np.int64),
+                                    np.zeros(n + 1, np.int64)),
+                                   shape=(n, n)))
+    check("zero-nnz spmv", float(torch.abs(Z @ x_g).sum()) == 0.0)
+    check("zero-nnz spgemm", (Z @ Z).nnz == 0)
+    check("zero-nnz transpose", Z.T.nnz == 0)
+    check("zero-nnz sum", float(Z.sum()) == 0.0)
+
     if runtime.rank == 0:
         print("DIST_ALL_OK", flush=True)
 
