@@ -25,6 +25,8 @@
 
 #include "common.h"
 
+#include <cstdlib>
+
 namespace {
 
 constexpr uint32_t EMPTY = 0xFFFFFFFFu;
@@ -455,6 +457,77 @@ __global__ __launch_bounds__(LS_THREADS) void spgemm_merge_numeric_kernel(
     }
   }
   if (COUNT && lane == 0) row_nnz_out[row] = out;
+}
+
+// Lane-per-row SCALAR merge for the shortest-A bin (a_len <= 8): all 8
+// merge heads live in registers and the k-way min / accumulate runs as
+// straight-line per-lane code — no shuffles, no cross-lane dependency
+// chains, 64 rows per wave.  The wave-merge kernel spends ~10 wave ops
+// per OUTPUT element on a grp_min→grp_sum latency chain; here the same
+// output costs unrolled VALU with 8-deep ILP.  Memory stays friendly on
+// banded/stencil matrices: adjacent lanes hold adjacent rows, whose
+// k-th B rows are adjacent in memory.  Selected by LS_SPGEMM_SCALAR.
+template <typename T, typename I, bool COUNT>
+__global__ __launch_bounds__(LS_THREADS) void spgemm_merge_scalar_kernel(
+    const i64* __restrict__ rows_list, i64 n_bin_rows,
+    const i64* __restrict__ A_indptr, const I* __restrict__ A_indices,
+    const T* __restrict__ A_vals, const i64* __restrict__ B_indptr,
+    const I* __restrict__ B_indices, const T* __restrict__ B_vals,
+    const i64* __restrict__ C_indptr, I* __restrict__ C_indices,
+    T* __restrict__ C_vals, i64* __restrict__ row_nnz_out) {
+  constexpr int MAXA = 8;
+  const i64 idx = (i64)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= n_bin_rows) return;
+  const i64 row = rows_list[idx];
+  const i64 s = A_indptr[row];
+  const int alen = (int)(A_indptr[row + 1] - s);
+  i64 kp[MAXA], kend[MAXA];
+  T a[MAXA], hval[MAXA];
+  uint32_t hcol[MAXA];
+#pragma unroll
+  for (int h = 0; h < MAXA; ++h) {
+    kp[h] = 0;
+    kend[h] = 0;
+    a[h] = ls_zero<T>();
+    hval[h] = ls_zero<T>();
+    hcol[h] = EMPTY;
+    if (h < alen) {
+      const i64 k = (i64)A_indices[s + h];
+      a[h] = A_vals[s + h];
+      kp[h] = B_indptr[k];
+      kend[h] = B_indptr[k + 1];
+      if (kp[h] < kend[h]) {
+        hcol[h] = (uint32_t)B_indices[kp[h]];
+        hval[h] = B_vals[kp[h]];
+      }
+    }
+  }
+  const i64 base = C_indptr[row];
+  i64 out = 0;
+  while (true) {
+    uint32_t mn = hcol[0];
+#pragma unroll
+    for (int h = 1; h < MAXA; ++h) mn = hcol[h] < mn ? hcol[h] : mn;
+    if (mn == EMPTY) break;
+    T acc = ls_zero<T>();
+#pragma unroll
+    for (int h = 0; h < MAXA; ++h) {
+      if (hcol[h] == mn) {
+        acc += a[h] * hval[h];
+        ++kp[h];
+        if (kp[h] < kend[h]) {
+          hcol[h] = (uint32_t)B_indices[kp[h]];
+          hval[h] = B_vals[kp[h]];
+        } else {
+          hcol[h] = EMPTY;
+        }
+      }
+    }
+    C_indices[base + out] = (I)mn;
+    C_vals[base + out] = acc;
+    ++out;
+  }
+  if (COUNT) row_nnz_out[row] = out;
 }
 
 // Compact-then-sort numeric kernel for wave-contained groups (W <= 64):
@@ -944,11 +1017,26 @@ void merge_symbolic_dispatch(int wcfg, const i64* rl, i64 n, const i64* ap,
   else L(std::integral_constant<int, 64>{});
 }
 
+inline bool scalar_merge_enabled() {
+  static const bool on = [] {
+    const char* e = std::getenv("LS_SPGEMM_SCALAR");
+    return e && e[0] == '1';
+  }();
+  return on;
+}
+
 template <typename T, typename I, bool COUNT>
 void merge_numeric_dispatch(int wcfg, const i64* rl, i64 n, const i64* ap,
                             const I* ai, const T* av, const i64* bp,
                             const I* bi, const T* bv, const i64* cp, I* ci,
                             T* cv, i64* rn, hipStream_t s) {
+  if (wcfg == 0 && scalar_merge_enabled()) {
+    hipLaunchKernelGGL((spgemm_merge_scalar_kernel<T, I, COUNT>),
+                       dim3((n + LS_THREADS - 1) / LS_THREADS),
+                       dim3(LS_THREADS), 0, s, rl, n, ap, ai, av, bp, bi,
+                       bv, cp, ci, cv, rn);
+    return;
+  }
   auto L = [&](auto wt) {
     constexpr int W = decltype(wt)::value;
     constexpr int ROWS = LS_THREADS / W;
