@@ -600,6 +600,9 @@ class csr_array(CompressedBase):
         self._global_nnz = None
         self._halo_cache = None
         self._ov_cache = None
+        # row lengths shrank: stale cache would be a (safe) over-estimate
+        # that can pin a less-unrolled SpMV tier
+        self._max_nnz_cache = None
         return self
 
     def sum_duplicates(self):
