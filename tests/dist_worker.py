@@ -154,6 +154,34 @@ def main():
     check("zero-nnz transpose", Z.T.nnz == 0)
     check("zero-nnz sum", float(Z.sum()) == 0.0)
 
+    # --- complex dtype end-to-end (SpMV + SpGEMM + transpose-conj) ---
+    rngc = np.random.default_rng(11)
+    Sc = sp.random(64, 64, 0.25, format="csr", random_state=12)
+    Sc = (Sc + 1j * sp.random(64, 64, 0.25, format="csr",
+                              random_state=13)).tocsr()
+    Ac = lsp.csr_array(Sc)
+    xc = rngc.random(64) + 1j * rngc.random(64)
+    yc = to_np(lsu.full_vector(Ac @ xc, 64))
+    check("complex spmv", np.allclose(yc, Sc @ xc))
+    Cc = Ac @ Ac
+    gotc = np.zeros((64, 64), dtype=np.complex128)
+    lo, hi = Cc.row_range
+    if hi > lo:
+        gotc[lo:hi] = to_np(Cc.todense())
+    tc = torch.from_numpy(gotc)
+    dist.all_reduce(tc)
+    check("complex spgemm", np.allclose(tc.numpy(), (Sc @ Sc).toarray(),
+                                        rtol=1e-12, atol=1e-12))
+    AH = Ac.conj().T
+    goth = np.zeros((64, 64), dtype=np.complex128)
+    lo, hi = AH.row_range
+    if hi > lo:
+        goth[lo:hi] = to_np(AH.todense())
+    th = torch.from_numpy(goth)
+    dist.all_reduce(th)
+    check("complex conj-transpose",
+          np.allclose(th.numpy(), Sc.conj().T.toarray()))
+
     if runtime.rank == 0:
         print("DIST_ALL_OK", flush=True)
 
