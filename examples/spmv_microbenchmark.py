@@ -28,6 +28,13 @@ def main():
     ap.add_argument("--nnz-per-row", type=int, default=11)
     ap.add_argument("-i", "--iters", type=int, default=100)
     ap.add_argument("-w", "--warmup", type=int, default=5)
+    ap.add_argument("--repartition", action="store_true",
+                    help="alternate A@x->y and A@y->x every iteration "
+                         "(the reference's mode that forces Legion to "
+                         "re-derive partitions each step, spmv_"
+                         "microbenchmark.py:18-29; here halo plans are "
+                         "cached per matrix, so the delta measures that "
+                         "the SPMD design has no repartitioning cost)")
     args = ap.parse_args()
 
     n = args.nmin
@@ -39,11 +46,17 @@ def main():
         lo, hi = runtime.local_range(n)
         x = torch.rand(hi - lo, dtype=torch.float64, device=runtime.device)
         y = torch.empty_like(x)
-        for _ in range(args.warmup):
-            lsp.spmv(A, x, y)
-        with Timer() as t:
-            for _ in range(args.iters):
+        def step(i):
+            if args.repartition and i % 2:
+                lsp.spmv(A, y, x)
+            else:
                 lsp.spmv(A, x, y)
+
+        for i in range(args.warmup):
+            step(i)
+        with Timer() as t:
+            for i in range(args.iters):
+                step(i)
         per = t.elapsed / args.iters
         nnz = A.nnz
         bpi = 4 if n < 2 ** 31 else 8  # index bytes
