@@ -84,11 +84,35 @@ def build_prolongation(nxf, nxc):
                      _local=(indptr, indices, data.contiguous()))
 
 
+def build_injection(nxf, nxc):
+    """Injection prolongation P = R^T for R[I, fine(2I+1)] = 1 — the
+    reference's default intergrid operator (gmg.py:201-212; on our
+    interior 2^k-1 grids coarse (I, J) sits at fine (2I+1, 2J+1)).
+    Cheaper transfers than bilinear but weaker coarse coupling (with a
+    5-pt operator the Galerkin product decouples to the diagonal), so
+    linear stays the default here."""
+    dev = runtime.device
+    nf = nxf * nxf
+    lo, hi = runtime.local_range(nf)
+    r = torch.arange(lo, hi, device=dev)
+    fi = torch.div(r, nxf, rounding_mode="floor")
+    fj = r % nxf
+    is_c = ((fi % 2) == 1) & ((fj % 2) == 1)
+    ci = torch.div(fi - 1, 2, rounding_mode="floor")
+    cj = torch.div(fj - 1, 2, rounding_mode="floor")
+    col = (ci * nxc + cj)[is_c]
+    data = torch.ones(col.numel(), dtype=torch.float64, device=dev)
+    indptr = torch.zeros(hi - lo + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(is_c.to(torch.int64), dim=0, out=indptr[1:])
+    return csr_array(None, shape=(nf, nxc * nxc),
+                     _local=(indptr, col, data))
+
+
 class GMG(LinearOperator):
     """V-cycle preconditioner as a LinearOperator for cg(M=...)."""
 
     def __init__(self, A, k, n_levels=None, nu=2, omega=0.8,
-                 coarse_iters=30):
+                 coarse_iters=30, gridop="linear"):
         self.nu = nu
         self.omega = omega
         self.coarse_iters = coarse_iters
@@ -104,7 +128,9 @@ class GMG(LinearOperator):
             if l == n_levels - 1:
                 break
             nxc = 2 ** (k - l - 1) - 1
-            P = build_prolongation(nx, nxc)
+            builder = {"linear": build_prolongation,
+                       "injection": build_injection}[gridop]
+            P = builder(nx, nxc)
             self.P.append(P)
             R = P.T
             Al = (R @ Al) @ P
@@ -196,6 +222,18 @@ class GMG(LinearOperator):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("-k", type=int, default=8, help="grid edge = 2^k - 1")
+    ap.add_argument("-d", "--data", choices=["poisson", "diffusion"],
+                    default="poisson",
+                    help="problem instance (reference gmg.py:430-438)")
+    ap.add_argument("-g", "--gridop", choices=["linear", "injection"],
+                    default="linear",
+                    help="intergrid transfer operator (reference "
+                         "gmg.py:448-456; linear converges in ~10 iters, "
+                         "injection is the reference's cheaper default)")
+    ap.add_argument("--epsilon", type=float, default=0.1,
+                    help="diffusion anisotropy (with -d diffusion)")
+    ap.add_argument("--theta", type=float, default=np.pi / 4,
+                    help="diffusion rotation angle (with -d diffusion)")
     ap.add_argument("--rtol", type=float, default=1e-10)
     ap.add_argument("--maxiter", type=int, default=200)
     ap.add_argument("--nu", type=int, default=2)
@@ -204,8 +242,12 @@ def main():
     nx = 2 ** args.k - 1
     from common import d2_mat_dirichlet_2d
     with Timer() as t_build:
-        A = d2_mat_dirichlet_2d(nx, nx)
-        M = GMG(A, args.k, nu=args.nu)
+        if args.data == "diffusion":
+            from legate_sparse.gallery import diffusion_2d
+            A = diffusion_2d(nx, args.epsilon, args.theta)
+        else:
+            A = d2_mat_dirichlet_2d(nx, nx)
+        M = GMG(A, args.k, nu=args.nu, gridop=args.gridop)
     n = A.shape[0]
     lo, hi = runtime.local_range(n)
     print0(f"build+hierarchy: {t_build.elapsed*1e3:.1f} ms  "

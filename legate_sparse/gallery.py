@@ -118,6 +118,81 @@ def poisson_2d(nx: int, ny: int, dtype=np.float64):
                      _local=(indptr, indices, data))
 
 
+def stencil_grid(S, grid, dtype=np.float64):
+    """Assemble the operator of a 2-D stencil over an nx*ny grid as a
+    distributed CSR (each rank builds only its local row block).
+
+    PyAMG gallery convention, which the reference's examples use
+    (reference examples/common.py:252-318 builds the same operator via
+    replicated DIA diagonals): grid point (i, j) (row-major) couples to
+    (i+di, j+dj) with coefficient ``S[c0+di, c1+dj]`` where (c0, c1) is
+    the stencil center; couplings leaving the grid are dropped
+    (homogeneous Dirichlet)."""
+    import torch
+    from .csr import csr_array
+    from .runtime import runtime
+
+    S = np.asarray(S)
+    if S.ndim != 2 or len(grid) != 2:
+        raise ValueError("stencil_grid supports 2-D stencils/grids")
+    if any(s % 2 == 0 for s in S.shape):
+        raise ValueError("stencil dimensions must be odd")
+    nx, ny = int(grid[0]), int(grid[1])
+    c0, c1 = S.shape[0] // 2, S.shape[1] // 2
+    tdt = to_torch_dtype(dtype)
+    M = nx * ny
+    lo, hi = runtime.local_range(M)
+    lr = hi - lo
+    dev = runtime.device
+    r = torch.arange(lo, hi, device=dev)
+    gi = torch.div(r, ny, rounding_mode="floor")
+    gj = r % ny
+    # nonzero offsets in ascending column-offset order -> sorted rows
+    offs = sorted(((di - c0, dj - c1, S[di, dj])
+                   for di in range(S.shape[0])
+                   for dj in range(S.shape[1]) if S[di, dj] != 0),
+                  key=lambda t: t[0] * ny + t[1])
+    cols, wts = [], []
+    for di, dj, v in offs:
+        ii, jj = gi + di, gj + dj
+        valid = (ii >= 0) & (ii < nx) & (jj >= 0) & (jj < ny)
+        cols.append(torch.where(valid, ii * ny + jj,
+                                torch.full_like(r, -1)))
+        wts.append(torch.where(
+            valid, torch.full((lr,), v.item(), dtype=tdt, device=dev),
+            torch.zeros(lr, dtype=tdt, device=dev)))
+    cols = torch.stack(cols, dim=1)
+    wts = torch.stack(wts, dim=1)
+    mask = cols >= 0
+    indices = cols[mask]
+    data = wts[mask].contiguous()
+    counts = mask.sum(dim=1)
+    indptr = torch.zeros(lr + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(counts, dim=0, out=indptr[1:])
+    return csr_array(None, shape=(M, M), _local=(indptr, indices, data))
+
+
+def diffusion_2d(N: int, epsilon: float = 1.0, theta: float = 0.0,
+                 dtype=np.float64):
+    """Rotated anisotropic diffusion operator on an N*N grid — the
+    reference's second GMG problem instance (examples/gmg.py -d
+    diffusion -> common.py:330-347).  Classical Q1 finite-element
+    9-point stencil for  -div(Q diag(1, eps) Q^T grad u)  with rotation
+    Q = rot(theta) (the PyAMG gallery formula)."""
+    C = np.cos(theta)
+    Sn = np.sin(theta)
+    CS = C * Sn
+    CC = C * C
+    SS = Sn * Sn
+    a = (-epsilon - 1) * CC + (-epsilon - 1) * SS + (3 * epsilon - 3) * CS
+    b = (2 * epsilon - 4) * CC + (-4 * epsilon + 2) * SS
+    c = (-epsilon - 1) * CC + (-epsilon - 1) * SS + (-3 * epsilon + 3) * CS
+    d = (-4 * epsilon + 2) * CC + (2 * epsilon - 4) * SS
+    e = (8 * epsilon + 8) * (CC + SS)
+    st = np.array([[a, b, c], [d, e, d], [c, b, a]]) / 6.0
+    return stencil_grid(st, (N, N), dtype=dtype)
+
+
 def rmat(scale: int, edge_factor: int = 16, seed: int = 0,
          dtype=np.float64):
     """Synthetic R-MAT (Graph500-style) adjacency matrix, built

@@ -89,3 +89,43 @@ def test_dia_ufuncs_and_astype():
     want = sp.dia_array((np.sqrt(data), offsets), shape=(6, 6)).tocsr()
     np.testing.assert_allclose(to_np(got.todense()), want.toarray())
     assert A.astype(np.float32).dtype == np.dtype(np.float32)
+
+
+def test_stencil_grid_matches_poisson():
+    """5-pt stencil through stencil_grid == the direct poisson builder
+    (transposed grid convention: stencil_grid is (rows, cols))."""
+    from legate_sparse.gallery import stencil_grid, poisson_2d
+    S5 = np.array([[0, -1, 0], [-1, 4, -1], [0, -1, 0]], dtype=float)
+    A = stencil_grid(S5, (9, 7))
+    B = poisson_2d(7, 9)
+    assert np.allclose(to_np(A.todense()), to_np(B.todense()))
+
+
+def test_stencil_grid_9pt_boundaries():
+    """9-pt stencil: interior row has 9 entries, corner row loses the
+    out-of-grid couplings; matches a dense construction."""
+    from legate_sparse.gallery import stencil_grid
+    rng = np.random.default_rng(3)
+    S = rng.standard_normal((3, 3))
+    nx, ny = 5, 6
+    A = to_np(stencil_grid(S, (nx, ny)).todense())
+    want = np.zeros((nx * ny, nx * ny))
+    for i in range(nx):
+        for j in range(ny):
+            for di in (-1, 0, 1):
+                for dj in (-1, 0, 1):
+                    ii, jj = i + di, j + dj
+                    if 0 <= ii < nx and 0 <= jj < ny:
+                        want[i * ny + j, ii * ny + jj] = S[di + 1, dj + 1]
+    assert np.allclose(A, want)
+
+
+def test_diffusion_2d_spd():
+    """Rotated anisotropic diffusion (reference examples/common.py:
+    330-347 instance): symmetric positive definite, zero interior row
+    sums (conservation)."""
+    from legate_sparse.gallery import diffusion_2d
+    D = to_np(diffusion_2d(8, 0.1, np.pi / 4).todense())
+    assert np.allclose(D, D.T)
+    assert np.linalg.eigvalsh(D).min() > 0
+    assert abs(D[3 * 8 + 4].sum()) < 1e-12

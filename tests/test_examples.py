@@ -33,6 +33,18 @@ def test_gmg_runs():
     assert "PASS" in out
 
 
+def test_gmg_diffusion():
+    """-d diffusion instance (reference gmg.py:430-438 parity)."""
+    out = _run("gmg.py", "-k", "5", "-d", "diffusion")
+    assert "PASS" in out
+
+
+def test_gmg_injection_gridop():
+    """--gridop injection (reference gmg.py:448-456 parity)."""
+    out = _run("gmg.py", "-k", "5", "-g", "injection", "--maxiter", "400")
+    assert "PASS" in out
+
+
 def test_spmv_microbenchmark():
     out = _run("spmv_microbenchmark.py", "--nmin", "4096", "--nmax", "8192",
                "-i", "3", "-w", "1")
