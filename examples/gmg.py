@@ -276,7 +276,13 @@ def main():
     print0(f"solve: {t.elapsed*1e3:.1f} ms in {n_it} iters "
            f"({t.elapsed/max(n_it,1)*1e3:.2f} ms/iter), info={info}, "
            f"relative residual {rel:.3e}")
-    assert rel <= max(args.rtol * 100, 1e-8), "residual check FAILED"
+    if rel > max(args.rtol * 100, 1e-8):
+        # injection transfers degrade with grid size on 5-pt operators
+        # (Galerkin R A R^T decouples toward the diagonal) — expected to
+        # need more iterations; report cleanly instead of tracebacking
+        print0(f"FAILED: residual {rel:.3e} above gate after {n_it} "
+               f"iters (try --gridop linear or a larger --maxiter)")
+        sys.exit(1)
     print0("PASS")
 
 

@@ -93,6 +93,15 @@ def test_gmg_gpu():
     assert "PASS" in out
 
 
+@pytest.mark.gpu
+def test_gmg_diffusion_gpu():
+    import torch
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    out = _run("gmg.py", "-k", "7", "-d", "diffusion")
+    assert "PASS" in out
+
+
 def test_pde_exact_mode():
     out = _run("pde.py", "-n", "48", "--exact")
     assert "PASS" in out
