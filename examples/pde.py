@@ -26,7 +26,14 @@ from common import Timer, d2_mat_dirichlet_2d, print0
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("-n", "--nx", type=int, default=512,
-                    help="grid edge (rows = nx*nx)")
+                    help="grid points along X")
+    ap.add_argument("-m", "--ny", type=int, default=None,
+                    help="grid points along Y (default: nx; reference "
+                         "pde.py:235-242)")
+    ap.add_argument("-p", "--plot", action="store_true",
+                    help="plot per-iteration residuals (reference "
+                         "pde.py:244-250)")
+    ap.add_argument("-f", "--plot-filename", default="pde_residuals.png")
     ap.add_argument("--rtol", type=float, default=1e-10)
     ap.add_argument("-i", "--max-iters", type=int, default=None)
     ap.add_argument("-w", "--warmup", type=int, default=5)
@@ -37,7 +44,10 @@ def main():
                          "(p = sin(pi x) sin(pi y), O(h^2) error)")
     args = ap.parse_args()
 
-    nx = ny = args.nx
+    nx = args.nx
+    ny = args.ny if args.ny is not None else args.nx
+    if args.exact and nx != ny:
+        raise SystemExit("--exact requires a square grid (nx == ny)")
     with Timer() as t_build:
         A = d2_mat_dirichlet_2d(nx, ny)
         n = A.shape[0]
@@ -91,8 +101,29 @@ def main():
                f"({flops/ (t.elapsed/iters) / 1e9:.1f} GFLOP/s)")
     else:
         maxiter = args.max_iters or 10 * nx
+        res_hist = []
+        callback = None
+        if args.plot:
+            def callback(xk):  # true residual per iteration (diagnostics)
+                rk = b - A @ xk
+                nk = torch.linalg.norm(rk) ** 2
+                if runtime.world_size > 1:
+                    from legate_sparse import comm
+                    comm.allreduce_(nk)
+                res_hist.append(float(torch.sqrt(nk)))
         with Timer() as t:
-            x, info = lsp.linalg.cg(A, b, rtol=args.rtol, maxiter=maxiter)
+            x, info = lsp.linalg.cg(A, b, rtol=args.rtol, maxiter=maxiter,
+                                    callback=callback)
+        if args.plot and runtime.rank == 0 and res_hist:
+            import matplotlib
+            matplotlib.use("Agg")
+            import matplotlib.pyplot as plt
+            plt.semilogy(res_hist)
+            plt.xlabel("iteration")
+            plt.ylabel("||b - A x||")
+            plt.title(f"CG residuals, {nx}x{ny} Poisson")
+            plt.savefig(args.plot_filename, dpi=120)
+            print0(f"residual plot -> {args.plot_filename}")
         r = b - A @ x
         num = torch.linalg.norm(r) ** 2
         den = torch.linalg.norm(b) ** 2
