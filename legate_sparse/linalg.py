@@ -212,6 +212,11 @@ def cg(A, b, x0=None, *, rtol: float = 1e-5, atol: float = 0.0,
                                                        dtype=dtype))
 
     bnrm2 = _gnorm(b)
+    # b = 0: the exact solution is x = 0; entering the loop would divide
+    # 0/0 in the fused axpby (scipy returns immediately too).  One host
+    # sync before the loop, not inside it.
+    if float(bnrm2.item()) == 0.0:
+        return torch.zeros_like(b), 0
     atol_t = torch.clamp(bnrm2 * rtol, min=atol)
 
     if x0 is None:

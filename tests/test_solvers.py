@@ -144,3 +144,35 @@ def test_sparse_norms():
     assert abs(norm(A) - spla.norm(S)) < 1e-12
     assert abs(norm(A, 1) - spla.norm(S, 1)) < 1e-12
     assert abs(norm(A, np.inf) - spla.norm(S, np.inf)) < 1e-12
+
+
+def test_cg_zero_rhs():
+    """b = 0 returns x = 0 immediately (scipy semantics) instead of
+    dividing 0/0 in the fused update."""
+    import scipy.sparse as sp
+    P = (50 * sp.identity(30) + sp.random(30, 30, 0.2,
+                                          random_state=1)).tocsr()
+    P = (0.5 * (P + P.T)).tocsr()
+    x, info = lsp.linalg.cg(lsp.csr_array(P), np.zeros(30), rtol=1e-10)
+    assert info == 0
+    assert float(torch.linalg.norm(x)) == 0.0
+
+
+def test_gmres_edge_restarts():
+    """restart > n, restart = 1, maxiter < restart all converge or stop
+    cleanly."""
+    import scipy.sparse as sp
+    rng = np.random.default_rng(0)
+    n = 40
+    S = (sp.random(n, n, 0.3, format="csr", random_state=1)
+         + 10 * sp.identity(n)).tocsr()
+    A = lsp.csr_array(S)
+    b = rng.random(n)
+    x, info = lsp.linalg.gmres(A, b, restart=1000, maxiter=2000,
+                               rtol=1e-12)
+    assert info == 0
+    assert np.linalg.norm(S @ to_np(x) - b) < 1e-8
+    x, info = lsp.linalg.gmres(A, b, restart=1, maxiter=500, rtol=1e-10)
+    assert info == 0
+    x, info = lsp.linalg.gmres(A, b, restart=30, maxiter=5, rtol=1e-14)
+    assert info in (0, 5)
