@@ -210,12 +210,26 @@ class csr_array(CompressedBase):
         part = runtime.partition(M)
         lo, hi = part.lo(runtime.rank), part.hi(runtime.rank)
         s, e = int(indptr[lo]), int(indptr[hi])
-        self._init_local(
-            (indptr[lo:hi + 1] - indptr[lo]).to(device=device,
-                                                dtype=_INDEX_DTYPE),
-            indices[s:e].to(device=device, dtype=_INDEX_DTYPE),
-            data[s:e].to(device),
-            (M, N))
+        l_indptr = (indptr[lo:hi + 1] - indptr[lo]).to(device=device,
+                                                       dtype=_INDEX_DTYPE)
+        l_indices = indices[s:e].to(device=device, dtype=_INDEX_DTYPE)
+        l_data = data[s:e].to(device)
+        # every kernel (and has_sorted_indices) assumes canonical rows;
+        # scipy leaves raw triples unsorted until sort_indices() — here
+        # canonical form is established at construction.  Detection is
+        # one vectorized pass; the sort runs only when needed.
+        if l_indices.numel() > 1:
+            lr = hi - lo
+            row_of = torch.repeat_interleave(
+                torch.arange(lr, device=device),
+                l_indptr[1:] - l_indptr[:-1])
+            key = row_of * N + l_indices
+            # strict increase == sorted AND duplicate-free; equal keys
+            # are duplicates to sum (canonical form)
+            if bool((key[1:] <= key[:-1]).any()):
+                l_indptr, l_indices, l_data = _assemble_local_rows(
+                    row_of, l_indices, l_data, lr, N, dedup=True)
+        self._init_local(l_indptr, l_indices, l_data, (M, N))
 
     def _from_global_coo(self, data, row, col, shape, tdt, device):
         if shape is None:

@@ -101,3 +101,25 @@ def test_issparse_helpers():
     A = lsp.csr_array((3, 3))
     assert lsp.issparse(A) and lsp.isspmatrix(A) and lsp.isspmatrix_csr(A)
     assert not lsp.isspmatrix_csr(np.ones(3))
+
+
+def test_csr_triple_unsorted_canonicalized():
+    """Raw CSR triples with unsorted columns or duplicates are
+    canonicalized at construction (every kernel assumes sorted,
+    duplicate-free rows; scipy defers to sort_indices/sum_duplicates)."""
+    data = np.array([1.0, 2.0, 3.0])
+    indices = np.array([2, 0, 1])
+    indptr = np.array([0, 2, 3])
+    A = lsp.csr_array((data, indices, indptr), shape=(2, 3))
+    assert A.indices.tolist() == [0, 2, 1]
+    S = sp.csr_array((data, indices, indptr), shape=(2, 3))
+    assert np.allclose(to_np(A.todense()), S.toarray())
+
+    d2 = np.array([1.0, 2.0, 5.0])
+    i2 = np.array([0, 0, 1])
+    p2 = np.array([0, 2, 3])
+    B = lsp.csr_array((d2, i2, p2), shape=(2, 2))
+    S2 = sp.csr_array((d2, i2, p2), shape=(2, 2))
+    S2.sum_duplicates()
+    assert B.nnz == S2.nnz
+    assert np.allclose(to_np(B.todense()), S2.toarray())
