@@ -45,6 +45,10 @@ class dia_array(CompressedBase):
         self._offsets = offsets.to(torch.int64).contiguous()
         if self._offsets.numel() != self._data.shape[0]:
             raise ValueError("number of offsets does not match data rows")
+        if self._offsets.numel() != torch.unique(self._offsets).numel():
+            # scipy rejects duplicate offsets; silently keeping them
+            # would produce non-canonical CSR rows on conversion
+            raise ValueError("offset array contains duplicate values")
 
     # -- properties -----------------------------------------------------
     @property

@@ -129,3 +129,10 @@ def test_diffusion_2d_spd():
     assert np.allclose(D, D.T)
     assert np.linalg.eigvalsh(D).min() > 0
     assert abs(D[3 * 8 + 4].sum()) < 1e-12
+
+
+def test_dia_duplicate_offsets_rejected():
+    """scipy parity: duplicate offsets raise (silently keeping them
+    would make non-canonical CSR rows on conversion)."""
+    with pytest.raises(ValueError):
+        lsp.dia_array((np.ones((2, 4)), np.array([0, 0])), shape=(4, 4))
