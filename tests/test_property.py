@@ -72,3 +72,35 @@ def test_transpose_matches_scipy(inp):
     S = sp.csr_array((vals, (rows, cols)), shape=(m, n))
     np.testing.assert_allclose(to_np(A.T.todense()), S.T.toarray(),
                                rtol=1e-12, atol=1e-12)
+
+
+@st.composite
+def spd_inputs(draw):
+    n = draw(st.integers(4, 48))
+    density = draw(st.floats(0.05, 0.5))
+    seed = draw(st.integers(0, 2 ** 31 - 1))
+    rng = np.random.default_rng(seed)
+    S = sp.random(n, n, density, format="csr",
+                  random_state=int(rng.integers(2 ** 31)))
+    S = (0.5 * (S + S.T) + n * sp.identity(n)).tocsr()
+    b = rng.standard_normal(n)
+    return S, b
+
+
+@settings(max_examples=25, deadline=None)
+@given(spd_inputs())
+def test_cg_random_spd(inp):
+    S, b = inp
+    x, info = lsp.linalg.cg(lsp.csr_array(S), b, rtol=1e-12, maxiter=2000)
+    assert info == 0
+    assert np.linalg.norm(S @ to_np(x) - b) <= 1e-8 * np.linalg.norm(b)
+
+
+@settings(max_examples=15, deadline=None)
+@given(spd_inputs())
+def test_gmres_random(inp):
+    S, b = inp
+    x, info = lsp.linalg.gmres(lsp.csr_array(S), b, rtol=1e-10,
+                               restart=25, maxiter=1500)
+    assert info == 0
+    assert np.linalg.norm(S @ to_np(x) - b) <= 1e-6 * np.linalg.norm(b)
