@@ -87,7 +87,7 @@ def spd_inputs(draw):
     return S, b
 
 
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=10, deadline=None)
 @given(spd_inputs())
 def test_cg_random_spd(inp):
     S, b = inp
@@ -96,7 +96,7 @@ def test_cg_random_spd(inp):
     assert np.linalg.norm(S @ to_np(x) - b) <= 1e-8 * np.linalg.norm(b)
 
 
-@settings(max_examples=15, deadline=None)
+@settings(max_examples=6, deadline=None)
 @given(spd_inputs())
 def test_gmres_random(inp):
     S, b = inp
